@@ -1,0 +1,171 @@
+// kernels_gemm.hip — local dense C = alpha*A*B + beta*C, Float64,
+// COLUMN-major (Julia layout), MFMA-tiled for gfx950 (CDNA4).
+//
+// Replaces the worker-side `localpart(A) * Bjk` of
+// /root/reference/src/linalg.jl:224 (the per-tile GEMM inside
+// _matmatmul!, linalg.jl:190-253).  The reference runs OpenBLAS dgemm on
+// the host; here the chunk lives in HBM and the GEMM runs on the MFMA
+// f64 pipe (v_mfma_f64_16x16x4_f64).
+//
+// Structure (v1): 128x128 block tile, BK=16, 4 waves (2x2), each wave a
+// 64x64 sub-tile = 4x4 fragments of 16x16, LDS-staged operands with an
+// 18-double row stride (conflict-spread for ds_read_b64), accumulate in
+// AGPR f64x4.  Sizes not multiples of the tile fall back to a naive
+// per-thread kernel (parity path for small chunks).
+#include "common.hpp"
+
+namespace da {
+
+typedef double f64x4 __attribute__((ext_vector_type(4)));
+typedef double f64x2 __attribute__((ext_vector_type(2)));
+
+#define BM 128
+#define BN 128
+#define BK 16
+#define LSTR 18   // LDS row stride in doubles (bank-spread padding)
+
+__global__ __launch_bounds__(256, 2)
+void gemm_f64_mfma(const double* __restrict__ A, const double* __restrict__ B,
+                   double* __restrict__ C, int64_t m, int64_t n, int64_t k,
+                   int64_t lda, int64_t ldb, int64_t ldc,
+                   double alpha, double beta) {
+    __shared__ double As[BM * LSTR];   // As[mm][kk] at mm*LSTR + kk
+    __shared__ double Bs[BN * LSTR];   // Bs[nn][kk] at nn*LSTR + kk
+
+    const int tid = threadIdx.x;
+    const int wave = tid >> 6;
+    const int lane = tid & 63;
+    const int l16 = lane & 15;
+    const int l4 = lane >> 4;
+    const int wr = (wave >> 1) * 64;   // wave row origin in tile
+    const int wc = (wave & 1) * 64;    // wave col origin
+
+    const int64_t bm = (int64_t)blockIdx.x * BM;
+    const int64_t bn = (int64_t)blockIdx.y * BN;
+
+    f64x4 acc[4][4];
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+#pragma unroll
+        for (int j = 0; j < 4; ++j) acc[i][j] = {0.0, 0.0, 0.0, 0.0};
+
+    const int64_t ktiles = k / BK;
+    for (int64_t kt = 0; kt < ktiles; ++kt) {
+        const int64_t k0 = kt * BK;
+        // Stage A tile (BM x BK): column c of the tile is contiguous in
+        // global memory (column-major).  1024 double2 pieces, 4/thread.
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+            int idx = tid + r * 256;          // 0..1023
+            int c = idx >> 6;                 // k-column 0..15
+            int row2 = (idx & 63) * 2;        // row pair
+            f64x2 v = *reinterpret_cast<const f64x2*>(
+                A + (k0 + c) * lda + bm + row2);
+            As[(row2 + 0) * LSTR + c] = v.x;
+            As[(row2 + 1) * LSTR + c] = v.y;
+        }
+        // Stage B tile (BK x BN): column n' has BK contiguous doubles.
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+            int idx = tid + r * 256;
+            int nn = idx >> 3;                // tile column 0..127
+            int k2 = (idx & 7) * 2;           // k pair
+            f64x2 v = *reinterpret_cast<const f64x2*>(
+                B + (bn + nn) * ldb + k0 + k2);
+            *reinterpret_cast<f64x2*>(&Bs[nn * LSTR + k2]) = v;
+        }
+        __syncthreads();
+#pragma unroll
+        for (int kk = 0; kk < 4; ++kk) {
+            double a[4], b[4];
+            const int kof = kk * 4 + l4;
+#pragma unroll
+            for (int i = 0; i < 4; ++i)
+                a[i] = As[(wr + i * 16 + l16) * LSTR + kof];
+#pragma unroll
+            for (int j = 0; j < 4; ++j)
+                b[j] = Bs[(wc + j * 16 + l16) * LSTR + kof];
+#pragma unroll
+            for (int i = 0; i < 4; ++i)
+#pragma unroll
+                for (int j = 0; j < 4; ++j)
+                    acc[i][j] = __builtin_amdgcn_mfma_f64_16x16x4f64(
+                        a[i], b[j], acc[i][j], 0, 0, 0);
+        }
+        __syncthreads();
+    }
+
+    // Epilogue: C/D fragment map col = lane&15, row = (lane>>4)*4 + j
+    // (shape-determined layout, dtype-independent on gfx950).
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+            int64_t col = bn + wc + j * 16 + l16;
+            int64_t row0 = bm + wr + i * 16 + l4 * 4;
+            double* cp = C + col * ldc + row0;
+#pragma unroll
+            for (int q = 0; q < 4; ++q) {
+                double v = alpha * acc[i][j][q];
+                cp[q] = (beta == 0.0) ? v : v + beta * cp[q];
+            }
+        }
+    }
+}
+
+// Naive fallback for arbitrary shapes (small parity chunks).
+__global__ void gemm_f64_naive(const double* __restrict__ A,
+                               const double* __restrict__ B,
+                               double* __restrict__ C,
+                               int64_t m, int64_t n, int64_t k,
+                               int64_t lda, int64_t ldb, int64_t ldc,
+                               double alpha, double beta) {
+    int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int64_t j = (int64_t)blockIdx.y * blockDim.y + threadIdx.y;
+    if (i >= m || j >= n) return;
+    double s = 0.0;
+    for (int64_t kk = 0; kk < k; ++kk)
+        s += A[kk * lda + i] * B[j * ldb + kk];
+    double v = alpha * s;
+    C[j * ldc + i] = (beta == 0.0) ? v : v + beta * C[j * ldc + i];
+}
+
+// beta-only scaling when k == 0 (fill!/rmul! branch of linalg.jl:232-240)
+__global__ void scale_c(double* __restrict__ C, int64_t m, int64_t n,
+                        int64_t ldc, double beta) {
+    int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int64_t j = (int64_t)blockIdx.y * blockDim.y + threadIdx.y;
+    if (i >= m || j >= n) return;
+    C[j * ldc + i] = (beta == 0.0) ? 0.0 : beta * C[j * ldc + i];
+}
+
+int launch_gemm_f64(void* Cv, const void* Av, const void* Bv,
+                    int64_t m, int64_t n, int64_t k,
+                    int64_t lda, int64_t ldb, int64_t ldc,
+                    double alpha, double beta, hipStream_t s) {
+    if (m < 0 || n < 0 || k < 0)
+        return set_err(-3, "da_gemm_f64: bad shape");
+    if (m == 0 || n == 0) return 0;
+    const double* A = (const double*)Av;
+    const double* B = (const double*)Bv;
+    double* C = (double*)Cv;
+    if (k == 0 || alpha == 0.0) {
+        dim3 t(64, 4), g((m + 63) / 64, (n + 3) / 4);
+        hipLaunchKernelGGL(scale_c, g, t, 0, s, C, m, n, ldc, beta);
+        DA_CHECK_HIP(hipGetLastError());
+        return 0;
+    }
+    if (m % BM == 0 && n % BN == 0 && k % BK == 0) {
+        dim3 g(m / BM, n / BN);
+        hipLaunchKernelGGL(gemm_f64_mfma, g, dim3(256), 0, s,
+                           A, B, C, m, n, k, lda, ldb, ldc, alpha, beta);
+    } else {
+        dim3 t(64, 4), g((m + 63) / 64, (n + 3) / 4);
+        hipLaunchKernelGGL(gemm_f64_naive, g, t, 0, s,
+                           A, B, C, m, n, k, lda, ldb, ldc, alpha, beta);
+    }
+    DA_CHECK_HIP(hipGetLastError());
+    return 0;
+}
+
+} // namespace da

@@ -1,0 +1,284 @@
+"""DArray — the distributed-array type, SPMD over one rank per MI355X.
+
+Mirrors the reference's type and constructor surface
+(/root/reference/src/darray.jl):
+  DArray{T,N,A}              darray.jl:25-55   -> DArray (localpart in HBM)
+  DArray(init, dims, ...)    darray.jl:76-174  -> DArray(init=..., ...)
+  dzeros/dones/dfill         darray.jl:468-494
+  drand/drandn               darray.jl:502-532 (philox per BASELINE.md)
+  distribute                 darray.jl:544-555
+  localpart/localindices     darray.jl:330-337, :394-400
+  close / d_closeall         darray.jl:46-49, core.jl:67-103
+  fill!/rand!                darray.jl:822-834
+
+Execution model: every rank runs the same program (SPMD); metadata
+(dims, cuts, indices) is computed identically everywhere; only the local
+chunk lives on this rank's GPU.  Collective ops must be called by all
+ranks.  There is no CPU fallback: all compute goes through
+libdarray_hip.so (fails loudly without a GPU).
+"""
+import numpy as np
+
+from . import _ffi, comm, geometry
+from ._ffi import check, lib
+from ._opcodes import DTYPES, NUMPY_DTYPES, DTYPE_SIZE, RAND_KINDS
+import ctypes
+
+_registry = {}
+_next_id = [0]
+
+
+def _auto_init():
+    if not comm.initialized():
+        comm.init()
+
+
+class DArray:
+    """Block-distributed dense array; localpart is an HBM chunk."""
+
+    def __init__(self, dims, dtype="f64", dist=None, init=None,
+                 _alloc=True):
+        _auto_init()
+        rank, nr = comm.rank_info()
+        dims = tuple(int(d) for d in dims)
+        if dist is None:
+            dist = geometry.defaultdist(dims, nr)
+        dist = tuple(int(c) for c in dist)
+        np_chunks = 1
+        for c in dist:
+            np_chunks *= c
+        if np_chunks > nr:
+            raise ValueError("dist %r needs %d ranks, have %d"
+                             % (dist, np_chunks, nr))
+        self.dims = dims
+        self.dist = dist
+        self.dtype = dtype
+        self.rank = rank
+        self.nranks = nr
+        self.nchunks = np_chunks
+        self.idxs, self.cuts = geometry.chunk_indices(dims, dist)
+        self.id = _next_id[0]
+        _next_id[0] += 1
+        self._chunk = None
+        if rank < np_chunks:
+            self.lidx = self.idxs[rank]
+            self.lshape = geometry.shape_of(self.lidx)
+            self.lnumel = geometry.nelems(self.lidx)
+        else:
+            self.lidx = tuple((0, 0) for _ in dims)
+            self.lshape = tuple(0 for _ in dims)
+            self.lnumel = 0
+        if _alloc:
+            p = ctypes.c_void_p()
+            check(lib.da_alloc(max(self.lnumel, 1) * DTYPE_SIZE[dtype],
+                               DTYPES[dtype], ctypes.byref(p)))
+            self._chunk = p
+            _registry[self.id] = self
+        if init is not None:
+            arr = init(self.lidx)
+            arr = np.asfortranarray(arr,
+                                    dtype=np.dtype(NUMPY_DTYPES[dtype]))
+            if tuple(arr.shape) != self.lshape:
+                raise ValueError("init returned shape %r, want %r"
+                                 % (arr.shape, self.lshape))
+            self.set_localpart(arr)
+
+    # ---- lifetime (darray.jl:46-49 -> core.jl:67-103) ----
+    def close(self):
+        if self._chunk is not None:
+            check(lib.da_free(self._chunk))
+            self._chunk = None
+            _registry.pop(self.id, None)
+
+    def __del__(self):
+        try:
+            if self._chunk is not None and lib is not None:
+                lib.da_free(self._chunk)
+                self._chunk = None
+                _registry.pop(self.id, None)
+        except Exception:
+            pass
+
+    def __enter__(self):
+        return self
+
+    def __exit__(self, *a):
+        self.close()
+
+    # ---- metadata ----
+    @property
+    def shape(self):
+        return self.dims
+
+    @property
+    def size(self):
+        n = 1
+        for d in self.dims:
+            n *= d
+        return n
+
+    @property
+    def ndims(self):
+        return len(self.dims)
+
+    def samedist(self, other):
+        return (self.dims == other.dims and self.dist == other.dist
+                and self.dtype == other.dtype)
+
+    def _ptr(self):
+        if self._chunk is None:
+            raise _ffi.DArrayError("use after close (darray id %d)" % self.id)
+        return self._chunk
+
+    # ---- localpart access (darray.jl:330-337, :394-400) ----
+    def localindices(self):
+        return self.lidx
+
+    def localpart(self):
+        """D2H copy of the local chunk as a numpy array (column-major)."""
+        out = np.empty(self.lshape,
+                       dtype=np.dtype(NUMPY_DTYPES[self.dtype]), order="F")
+        if self.lnumel:
+            check(lib.da_d2h(self._ptr(), out.ctypes.data_as(ctypes.c_void_p),
+                             self.lnumel * DTYPE_SIZE[self.dtype]))
+        return out
+
+    def set_localpart(self, arr):
+        """H2D upload into the local chunk (d[:L] = v, darray.jl:378-382)."""
+        arr = np.asfortranarray(arr, dtype=np.dtype(NUMPY_DTYPES[self.dtype]))
+        if tuple(arr.shape) != self.lshape:
+            raise ValueError("shape mismatch")
+        if self.lnumel:
+            check(lib.da_h2d(self._ptr(),
+                             arr.ctypes.data_as(ctypes.c_void_p),
+                             self.lnumel * DTYPE_SIZE[self.dtype]))
+        return self
+
+    # ---- device-side content ops (darray.jl:822-834) ----
+    def fill_(self, v):
+        if self.lnumel:
+            check(lib.da_fill(self._ptr(), float(v), self.lnumel,
+                              DTYPES[self.dtype]))
+        return self
+
+    def rand_(self, kind="uniform", seed_base=1234):
+        """Per-rank philox seed = seed_base + rank (BASELINE.md protocol,
+        mirroring test/runtests.jl:23)."""
+        if self.lnumel:
+            check(lib.da_rand(self._ptr(), self.lnumel, DTYPES[self.dtype],
+                              seed_base + self.rank, RAND_KINDS[kind], 0))
+        return self
+
+    def similar(self, dtype=None):
+        return DArray(self.dims, dtype or self.dtype, self.dist)
+
+    def copy(self):
+        out = self.similar()
+        if self.lnumel:
+            check(lib.da_d2d(out._ptr(), self._ptr(),
+                             self.lnumel * DTYPE_SIZE[self.dtype]))
+        return out
+
+    # ---- conversion (Array(::DArray); collect) ----
+    def collect(self):
+        """Full array on every rank.  Cross-rank chunk movement is
+        control-plane (torch.distributed gloo when world>1), mirroring
+        the reference's remotecall gather (darray.jl:574ff)."""
+        npdt = np.dtype(NUMPY_DTYPES[self.dtype])
+        local = self.localpart()
+        if self.nchunks == 1:
+            out = np.zeros(self.dims, dtype=npdt, order="F")
+            if self.lnumel:
+                sl = tuple(slice(lo, hi) for lo, hi in self.lidx)
+                out[sl] = local
+            return out
+        import torch.distributed as td
+        if not td.is_initialized():
+            raise _ffi.DArrayError(
+                "collect() with nranks>1 needs torch.distributed (gloo) "
+                "initialized for the control plane")
+        gathered = [None] * self.nranks
+        td.all_gather_object(gathered, local)
+        out = np.zeros(self.dims, dtype=npdt, order="F")
+        for r in range(self.nchunks):
+            sl = tuple(slice(lo, hi) for lo, hi in self.idxs[r])
+            out[sl] = gathered[r]
+        return out
+
+    def __eq__(self, other):
+        if isinstance(other, DArray):
+            if self.dims != other.dims:
+                return False
+            a, b = self.localpart(), other_aligned_localpart(self, other)
+            import torch.distributed as td
+            same = bool(np.array_equal(a, b))
+            if self.nranks > 1 and td.is_initialized():
+                flags = [None] * self.nranks
+                td.all_gather_object(flags, same)
+                return all(flags)
+            return same
+        return NotImplemented
+
+    def __repr__(self):
+        return ("DArray(dims=%r, dist=%r, dtype=%s, rank=%d/%d, lshape=%r)"
+                % (self.dims, self.dist, self.dtype, self.rank, self.nranks,
+                   self.lshape))
+
+
+def other_aligned_localpart(d, other):
+    if d.dist != other.dist or d.dims != other.dims:
+        raise _ffi.DArrayError("== needs aligned distributions (round 1)")
+    return other.localpart()
+
+
+# ---- convenience constructors (darray.jl:460-532) ----
+def dzeros(dims, dtype="f64", dist=None):
+    return DArray(dims, dtype, dist).fill_(0.0)
+
+
+def dones(dims, dtype="f64", dist=None):
+    return DArray(dims, dtype, dist).fill_(1.0)
+
+
+def dfill(v, dims, dtype="f64", dist=None):
+    return DArray(dims, dtype, dist).fill_(v)
+
+
+def drand(dims, dtype="f64", dist=None, seed_base=1234):
+    return DArray(dims, dtype, dist).rand_("uniform", seed_base)
+
+
+def drandn(dims, dtype="f64", dist=None, seed_base=1234):
+    return DArray(dims, dtype, dist).rand_("normal", seed_base)
+
+
+def distribute(a, dist=None):
+    """Local (replicated) array -> DArray; each rank uploads its own
+    slice — the SPMD analog of the DestinationSerializer scatter
+    (darray.jl:544-555): only the local slice touches the wire/PCIe."""
+    a = np.asarray(a)
+    dtmap = {np.dtype("float64"): "f64", np.dtype("float32"): "f32",
+             np.dtype("int64"): "i64"}
+    dt = dtmap.get(a.dtype)
+    if dt is None:
+        raise ValueError("unsupported dtype %s" % a.dtype)
+    return DArray(a.shape, dt, dist,
+                  init=lambda idx: a[tuple(slice(lo, hi) for lo, hi in idx)])
+
+
+def localpart(d):
+    return d.localpart()
+
+
+def localindices(d):
+    return d.localindices()
+
+
+def d_closeall():
+    """core.jl:98-103."""
+    for d in list(_registry.values()):
+        d.close()
+
+
+def bytes_in_use():
+    return int(lib.da_bytes_in_use())

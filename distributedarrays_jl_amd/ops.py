@@ -1,0 +1,331 @@
+"""Hot-path operations on DArrays — the API mirror of
+src/mapreduce.jl, src/broadcast.jl and src/linalg.jl, executing HIP
+kernels on local chunks and RCCL collectives across ranks.
+
+The reference fans a closure out to each worker and runs Base on
+localparts; here every rank IS a worker (SPMD), the closure set is the
+fixed opcode table (include/darray_hip.h), and the caller-side fold
+becomes an RCCL allreduce (SURVEY.md §3.1 build mapping)."""
+import ctypes
+import numpy as np
+
+from . import comm, geometry, plan
+from ._ffi import check, lib, DArrayError
+from ._opcodes import (DTYPES, DTYPE_SIZE, NUMPY_DTYPES, MAP_OP, MAP2_OP,
+                       RED_OPS, RED_FS, I64_MAP_OPS, I64_MAP2_OPS)
+from .darray import DArray
+
+
+def _aligned(*ds):
+    d0 = ds[0]
+    for d in ds[1:]:
+        if d.dims != d0.dims or d.dist != d0.dist or d.dtype != d0.dtype:
+            raise DArrayError(
+                "operands must share dims/dist/dtype (aligned cuts); "
+                "mismatched-cuts makelocal is a next-tier row, SURVEY §8f")
+    return d0
+
+
+# ------------------------------------------------------- map / broadcast
+def map_(op, dest, src):
+    """map!(f, dest, src) — mapreduce.jl:5-12 (aligned cuts)."""
+    _aligned(dest, src)
+    if dest.dtype == "i64" and op not in I64_MAP_OPS:
+        raise DArrayError("op %r invalid for i64" % op)
+    if dest.lnumel:
+        check(lib.da_map(MAP_OP[op], dest._ptr(), src._ptr(),
+                         dest.lnumel, DTYPES[dest.dtype]))
+    return dest
+
+
+def dmap(op, src):
+    """map(f, d) — mapreduce.jl:3 (out-of-place)."""
+    return map_(op, src.similar(), src)
+
+
+def map2_(op, dest, a, b):
+    """dest .= f.(a, b) for the binary table (mapreduce.jl:180-189)."""
+    _aligned(dest, a, b)
+    if dest.dtype == "i64" and op not in I64_MAP2_OPS:
+        raise DArrayError("op %r invalid for i64" % op)
+    if dest.lnumel:
+        check(lib.da_map2(MAP2_OP[op], dest._ptr(), a._ptr(), b._ptr(),
+                          dest.lnumel, DTYPES[dest.dtype]))
+    return dest
+
+
+def elementwise(op, a, b):
+    return map2_(op, a.similar(), a, b)
+
+
+def broadcast_fma(dest, a, b, c):
+    """D .= A .* B .+ c — the fused cfg-3 broadcast (broadcast.jl:65-85;
+    aligned same-cuts args need zero communication)."""
+    _aligned(dest, a, b)
+    if dest.lnumel:
+        check(lib.da_bcast_fma(dest._ptr(), a._ptr(), b._ptr(), float(c),
+                               dest.lnumel, DTYPES[dest.dtype]))
+    return dest
+
+
+def axpy_(alpha, x, y):
+    """axpy! — linalg.jl:24-34."""
+    _aligned(x, y)
+    if y.lnumel:
+        check(lib.da_axpby(y._ptr(), x._ptr(), float(alpha), 1.0,
+                           y.lnumel, DTYPES[y.dtype]))
+    return y
+
+
+def add_(dest, src, scale=1.0):
+    """add! — linalg.jl:62-76."""
+    _aligned(dest, src)
+    if dest.lnumel:
+        check(lib.da_add(dest._ptr(), src._ptr(), float(scale),
+                         dest.lnumel, DTYPES[dest.dtype]))
+    return dest
+
+
+def scale_(a, s):
+    """rmul! — linalg.jl:54-59."""
+    if a.lnumel:
+        check(lib.da_scale(a._ptr(), float(s), a.lnumel, DTYPES[a.dtype]))
+    return a
+
+
+# ------------------------------------------------------------ reductions
+_CTYPE = {"f64": ctypes.c_double, "f32": ctypes.c_float,
+          "i64": ctypes.c_int64}
+
+
+def _local_reduce(f, op, d):
+    out = _CTYPE[d.dtype]()
+    check(lib.da_reduce(RED_FS[f], RED_OPS[op], d._ptr(), d.lnumel,
+                        DTYPES[d.dtype], ctypes.byref(out)))
+    return out
+
+
+def mapreduce(f, op, d):
+    """mapreduce(f, op, d) — mapreduce.jl:29-35: per-chunk kernel
+    reduction, then the cross-worker fold as one scalar RCCL allreduce
+    (tree; partials are 8 bytes — latency-bound)."""
+    out = _local_reduce(f, op, d)
+    if d.nranks > 1:
+        check(lib.da_allreduce(ctypes.byref(out), 1, DTYPES[d.dtype],
+                               RED_OPS[op]))
+    return out.value
+
+
+def dsum(d):
+    return mapreduce("identity", "add", d)
+
+
+def dprod(d):
+    return mapreduce("identity", "mul", d)
+
+
+def dmaximum(d):
+    if d.size == 0:
+        raise DArrayError("maximum over empty DArray")  # Julia throws too
+    return mapreduce("identity", "max", d)
+
+
+def dminimum(d):
+    if d.size == 0:
+        raise DArrayError("minimum over empty DArray")
+    return mapreduce("identity", "min", d)
+
+
+def dextrema(d):
+    """extrema — mapreduce.jl:124-131."""
+    return (dminimum(d), dmaximum(d))
+
+
+def dmean(d):
+    """mean rides the sum path (ext/StatisticsExt.jl:6)."""
+    return dsum(d) / d.size
+
+
+def ddot(x, y):
+    """dot — linalg.jl:36-45 (aligned cuts: local fused path)."""
+    _aligned(x, y)
+    tmp = elementwise("mul", x, y)
+    try:
+        return mapreduce("identity", "add", tmp)
+    finally:
+        tmp.close()
+
+
+def dnorm(x, p=2):
+    """norm — linalg.jl:47-52 (p in {1, 2, inf})."""
+    if p == 2:
+        return float(np.sqrt(mapreduce("abs2", "add", x)))
+    if p == 1:
+        return mapreduce("abs", "add", x)
+    if p == float("inf"):
+        return mapreduce("abs", "max", x)
+    raise DArrayError("norm: p=%r not supported" % p)
+
+
+# --------------------------------------------------------------- matmul
+class _Buf:
+    """Raw device staging buffer."""
+
+    def __init__(self, nbytes):
+        self.p = ctypes.c_void_p()
+        check(lib.da_alloc(max(int(nbytes), 1), 0, ctypes.byref(self.p)))
+
+    def at(self, byte_off):
+        return ctypes.c_void_p(self.p.value + byte_off)
+
+    def free(self):
+        if self.p.value:
+            check(lib.da_free(self.p))
+            self.p = ctypes.c_void_p()
+
+
+def _copy2d(dst, dpitch, src, spitch, width, height):
+    if width and height:
+        check(lib.da_copy2d(dst, dpitch, src, spitch, width, height))
+
+
+def dmatmul(A, B, alpha=1.0):
+    """C = alpha * A * B — the `*` wrapper (linalg.jl:266-273) plus
+    _matmatmul! (linalg.jl:190-253), re-expressed per plan.py.  f64 only
+    (the cfg-4 metric path); local GEMM is the MFMA kernel."""
+    if A.dtype != "f64" or B.dtype != "f64":
+        raise DArrayError("dmatmul: f64 only (metric path)")
+    if A.ndims != 2 or B.ndims != 2 or A.dims[1] != B.dims[0]:
+        raise DArrayError("dmatmul: shape mismatch %r x %r"
+                          % (A.dims, B.dims))
+    m, kk = A.dims
+    n = B.dims[1]
+    I, J = A.dist
+    K = plan.c_grid(A.dist, B.dist)[1]
+    C = DArray((m, n), "f64", (I, K))
+    C.fill_(0.0)
+    r = A.rank
+    pos = plan.a_rank_pos(r, A.dist)
+    esz = 8
+
+    pieces = plan.bslab_plan(A.dist, A.cuts[1], B.dims, B.dist, B.idxs)
+    my_sends = [p for p in pieces if p[0] == r and p[1] != r]
+    my_recvs = [p for p in pieces if p[1] == r and p[0] != r]
+    my_local = [p for p in pieces if p[0] == r and p[1] == r]
+
+    slab = None
+    sendbufs, recvbufs = [], []
+    if pos is not None:
+        i, j = pos
+        rlo, rhi = plan.slab_rows(A.cuts[1], j)
+        srows = rhi - rlo
+        slab = _Buf(srows * n * esz)
+
+        # pack my outgoing pieces from my B block (da_copy2d: column-major
+        # sub-block -> contiguous)
+        if my_sends or my_recvs or my_local:
+            Brows = B.lshape[0] if B.lnumel else 0
+            for (src, dst, rows, cols) in my_sends:
+                prows = rows[1] - rows[0]
+                pcols = cols[1] - cols[0]
+                buf = _Buf(prows * pcols * esz)
+                off = ((rows[0] - B.lidx[0][0])
+                       + (cols[0] - B.lidx[1][0]) * Brows) * esz
+                _copy2d(buf.p, prows * esz,
+                        ctypes.c_void_p(B._ptr().value + off), Brows * esz,
+                        prows * esz, pcols)
+                sendbufs.append(((src, dst, rows, cols), buf))
+            for (src, dst, rows, cols) in my_recvs:
+                prows = rows[1] - rows[0]
+                pcols = cols[1] - cols[0]
+                recvbufs.append(((src, dst, rows, cols),
+                                 _Buf(prows * pcols * esz)))
+            # exchange (grouped so send/recv pairs match globally)
+            if my_sends or my_recvs:
+                check(lib.da_group_start())
+                for (src, dst, rows, cols), buf in sendbufs:
+                    nb = (rows[1] - rows[0]) * (cols[1] - cols[0]) * esz
+                    check(lib.da_send(buf.p, nb, dst))
+                for (src, dst, rows, cols), buf in recvbufs:
+                    nb = (rows[1] - rows[0]) * (cols[1] - cols[0]) * esz
+                    check(lib.da_recv(buf.p, nb, src))
+                check(lib.da_group_end())
+            # unpack into the slab (stream-ordered after the group)
+            for (src, dst, rows, cols) in my_local:
+                prows = rows[1] - rows[0]
+                pcols = cols[1] - cols[0]
+                off = ((rows[0] - B.lidx[0][0])
+                       + (cols[0] - B.lidx[1][0]) * (B.lshape[0])) * esz
+                doff = ((rows[0] - rlo) + cols[0] * srows) * esz
+                _copy2d(slab.at(doff), srows * esz,
+                        ctypes.c_void_p(B._ptr().value + off),
+                        B.lshape[0] * esz, prows * esz, pcols)
+            for (src, dst, rows, cols), buf in recvbufs:
+                prows = rows[1] - rows[0]
+                pcols = cols[1] - cols[0]
+                doff = ((rows[0] - rlo) + cols[0] * srows) * esz
+                _copy2d(slab.at(doff), srows * esz, buf.p, prows * esz,
+                        prows * esz, pcols)
+    elif my_sends or my_recvs or my_local:
+        raise DArrayError("B owner outside A's grid unsupported (round 1)")
+
+    # local partial GEMMs: partial[k] = A_local @ slab[:, ccols[k]]
+    ccols = geometry.ranges1d(C.cuts[1])
+    partials = []
+    if pos is not None and A.lnumel:
+        i, j = pos
+        mloc = A.lshape[0]
+        kloc = A.lshape[1]
+        for k in range(K):
+            clo, chi = ccols[k]
+            nk = chi - clo
+            pk = _Buf(mloc * nk * esz)
+            check(lib.da_gemm_f64(pk.p, A._ptr(), slab.at(clo * kloc * esz),
+                                  mloc, nk, kloc, mloc, kloc, mloc,
+                                  1.0, 0.0))
+            partials.append(pk)
+
+    # partial exchange + ordered accumulation (linalg.jl:243-251)
+    moves = plan.partial_plan(A.dist, K)
+    my_psends = [mv for mv in moves if mv[0] == r]
+    my_precvs = [mv for mv in moves if mv[1] == r]
+    precv = {}
+    if my_psends or my_precvs:
+        for (src, dst, k) in my_precvs:
+            srci, srcj = src % I, src // I
+            mloc = C.lshape[0]
+            nk = C.lshape[1]
+            precv[(src, k)] = _Buf(mloc * nk * esz)
+        check(lib.da_group_start())
+        for (src, dst, k) in my_psends:
+            nb = A.lshape[0] * (ccols[k][1] - ccols[k][0]) * esz
+            check(lib.da_send(partials[k].p, nb, dst))
+        for (src, dst, k) in my_precvs:
+            buf = precv[(src, k)]
+            nb = C.lshape[0] * C.lshape[1] * esz
+            check(lib.da_recv(buf.p, nb, src))
+        check(lib.da_group_end())
+    if r < I * K and C.lnumel:
+        i, myk = r % I, r // I
+        for j in plan.accumulate_order(J):
+            srcrank = i + I * j
+            if srcrank == r:
+                check(lib.da_add(C._ptr(), partials[myk].p, float(alpha),
+                                 C.lnumel, DTYPES["f64"]))
+            else:
+                buf = precv[(srcrank, myk)]
+                check(lib.da_add(C._ptr(), buf.p, float(alpha),
+                                 C.lnumel, DTYPES["f64"]))
+
+    check(lib.da_synchronize())
+    for _, b in sendbufs:
+        b.free()
+    for _, b in recvbufs:
+        b.free()
+    for b in precv.values():
+        b.free()
+    for b in partials:
+        b.free()
+    if slab is not None:
+        slab.free()
+    return C
