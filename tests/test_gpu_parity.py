@@ -1,0 +1,366 @@
+"""GPU parity tests: the HIP path vs the CPU oracle on identical seeded
+inputs (the reference's own differential-test pattern,
+/root/reference/test/runtests.jl + test/darray.jl — GPU DArray vs oracle
+instead of DArray vs Base Array).
+
+Exactness split (SURVEY.md §8c / BASELINE.json north_star):
+  bit-exact: philox fills, integer ops/reductions, IEEE-exact float ops
+             (+,-,*,/, sqrt, abs, neg, floor/ceil/trunc/rint, fma-free
+             a*b+c chains — all kernels compile -ffp-contract=off)
+  tolerance: transcendentals (OCML vs libm, few ulp), float reductions
+             and matmul (fold order; contract 1e-6 rel, tested tighter)
+"""
+import ctypes
+
+import numpy as np
+import pytest
+
+pytestmark = pytest.mark.gpu
+
+
+@pytest.fixture(scope="module")
+def dja():
+    import distributedarrays_jl_amd as dja
+    dja.comm.init()
+    yield dja
+    dja.d_closeall()
+
+
+from oracle import philox, ops as oops, geometry as ogeo
+
+# ops whose GPU result must be bit-identical to the numpy oracle
+EXACT_OPS = ["identity", "neg", "abs", "abs2", "inv", "sqrt", "floor",
+             "ceil", "round", "trunc", "sign"]
+# transcendental: few-ulp tolerance
+TRANS_OPS = ["cbrt", "exp", "exp2", "exp10", "expm1", "log", "log2",
+             "log10", "log1p", "sin", "cos", "tan", "asin", "acos",
+             "atan", "sinh", "cosh", "tanh", "asinh", "atanh",
+             "sinpi", "cospi", "deg2rad", "rad2deg", "sec", "csc", "cot"]
+# need |x| > 1 domain
+GT1_OPS = ["acosh"]
+
+
+# ------------------------------------------------------------------ fills
+@pytest.mark.parametrize("n", [1, 2, 3, 255, 4096, 1 << 20, (1 << 20) + 7])
+def test_drand_f64_bitexact(dja, n):
+    d = dja.drand((n,), "f64")
+    ref = philox.fill_uniform_f64(n, seed=1234)
+    assert np.array_equal(d.localpart(), ref)
+    d.close()
+
+
+def test_drand_f32_i64_bitexact(dja):
+    n = (1 << 18) + 5
+    d = dja.drand((n,), "f32")
+    assert np.array_equal(d.localpart(), philox.fill_uniform_f32(n, 1234))
+    d.close()
+    di = dja.drand((n,), "i64")
+    assert np.array_equal(di.localpart(), philox.fill_int64(n, 1234))
+    di.close()
+
+
+def test_drandn_tolerance_and_moments(dja):
+    n = 1 << 20
+    d = dja.drandn((n,), "f64")
+    got = d.localpart()
+    ref = philox.fill_normal_f64(n, seed=1234)
+    assert np.allclose(got, ref, rtol=1e-12, atol=1e-12)
+    assert abs(got.mean()) < 0.01 and abs(got.std() - 1) < 0.01
+    d.close()
+
+
+def test_fill(dja):
+    for dt, v in [("f64", 2.5), ("f32", -1.0), ("i64", 7)]:
+        d = dja.dfill(v, (1001,), dt)
+        assert (d.localpart() == v).all()
+        d.close()
+    z = dja.dzeros((64, 32))
+    assert (z.localpart() == 0).all()
+    z.close()
+
+
+# -------------------------------------------------------------------- map
+def _input_for(op, n=100003):
+    x = philox.fill_uniform_f64(n, seed=9)
+    if op in GT1_OPS:
+        return x + 1.5
+    if op in ("asin", "acos", "atanh"):
+        return x * 0.99
+    if op in ("log", "log2", "log10", "sqrt", "inv", "csc", "cot"):
+        return x + 0.01
+    return x
+
+
+@pytest.mark.parametrize("op", EXACT_OPS)
+def test_map_exact(dja, op):
+    x = _input_for(op)
+    d = dja.distribute(x)
+    out = dja.dmap(op, d)
+    ref = oops.oracle_map(op, x)
+    assert np.array_equal(out.localpart(), ref), op
+    out.close(); d.close()
+
+
+@pytest.mark.parametrize("op", TRANS_OPS + GT1_OPS)
+def test_map_transcendental(dja, op):
+    x = _input_for(op)
+    d = dja.distribute(x)
+    out = dja.dmap(op, d)
+    ref = oops.oracle_map(op, x)
+    assert np.allclose(out.localpart(), ref, rtol=1e-13, atol=1e-14), op
+    out.close(); d.close()
+
+
+def test_map_f32(dja):
+    x = philox.fill_uniform_f32(65537, seed=3)
+    d = dja.distribute(x)
+    out = dja.dmap("sin", d)
+    assert np.allclose(out.localpart(), np.sin(x), rtol=1e-6, atol=1e-7)
+    out.close(); d.close()
+
+
+def test_map_inplace(dja):
+    x = philox.fill_uniform_f64(4097, seed=4)
+    d = dja.distribute(x)
+    dja.map_("abs2", d, d)
+    assert np.array_equal(d.localpart(), x * x)
+    d.close()
+
+
+def test_map_i64(dja):
+    x = philox.fill_int64(10001, seed=5)
+    d = dja.distribute(x)
+    for op in ("identity", "neg", "abs", "abs2", "sign"):
+        out = dja.dmap(op, d)
+        with np.errstate(over="ignore"):
+            ref = {"identity": lambda v: v, "neg": lambda v: -v,
+                   "abs": np.abs, "abs2": lambda v: v * v,
+                   "sign": np.sign}[op](x)
+        assert np.array_equal(out.localpart(), ref), op
+        out.close()
+    d.close()
+
+
+# ------------------------------------------------------------------- map2
+def test_map2_exact(dja):
+    a = philox.fill_uniform_f64(50001, seed=11)
+    b = philox.fill_uniform_f64(50001, seed=12) + 0.5
+    da_, db = dja.distribute(a), dja.distribute(b)
+    for op in ("add", "sub", "mul", "div", "min2", "max2"):
+        out = dja.elementwise(op, da_, db)
+        assert np.array_equal(out.localpart(), oops.oracle_map2(op, a, b)), op
+        out.close()
+    for op in ("pow", "atan2", "mod", "rem"):
+        out = dja.elementwise(op, da_, db)
+        assert np.allclose(out.localpart(), oops.oracle_map2(op, a, b),
+                           rtol=1e-13), op
+        out.close()
+    da_.close(); db.close()
+
+
+def test_map2_i64(dja):
+    with np.errstate(over="ignore"):
+        a = philox.fill_int64(20001, seed=13)
+        b = np.abs(philox.fill_int64(20001, seed=14)) % 1000 + 1
+        da_, db = dja.distribute(a), dja.distribute(b)
+        for op in ("add", "sub", "mul", "idiv", "mod", "rem", "and", "or",
+                   "xor", "min2", "max2"):
+            out = dja.elementwise(op, da_, db)
+            ref = oops.oracle_map2(op, a, b)
+            assert np.array_equal(out.localpart(), ref), op
+            out.close()
+        da_.close(); db.close()
+
+
+def test_minmax_nan_propagation(dja):
+    a = philox.fill_uniform_f64(1000, seed=15)
+    a[137] = np.nan
+    d = dja.distribute(a)
+    assert np.isnan(dja.dmaximum(d))
+    assert np.isnan(dja.dminimum(d))
+    d.close()
+
+
+# -------------------------------------------------- fused broadcast / blas1
+def test_bcast_fma_bitexact(dja):
+    n = (1 << 20) + 3
+    a = philox.fill_uniform_f64(n, seed=21)
+    b = philox.fill_uniform_f64(n, seed=22)
+    da_, db = dja.distribute(a), dja.distribute(b)
+    dd = dja.dzeros((n,))
+    dja.broadcast_fma(dd, da_, db, 0.25)
+    assert np.array_equal(dd.localpart(), oops.oracle_bcast_fma(a, b, 0.25))
+    dd.close(); da_.close(); db.close()
+
+
+def test_axpy_add_scale_bitexact(dja):
+    n = 30011
+    x = philox.fill_uniform_f64(n, seed=23)
+    y = philox.fill_uniform_f64(n, seed=24)
+    dx, dy = dja.distribute(x), dja.distribute(y)
+    dja.axpy_(2.5, dx, dy)
+    ref = oops.oracle_axpy(2.5, x, y)
+    assert np.array_equal(dy.localpart(), ref)
+    dja.add_(dy, dx, 1.0)
+    ref = oops.oracle_add(ref, x, 1.0)
+    assert np.array_equal(dy.localpart(), ref)
+    dja.add_(dy, dx, -0.5)
+    ref = oops.oracle_add(ref, x, -0.5)
+    assert np.array_equal(dy.localpart(), ref)
+    dja.scale_(dy, 3.0)
+    ref = oops.oracle_scale(ref, 3.0)
+    assert np.array_equal(dy.localpart(), ref)
+    dx.close(); dy.close()
+
+
+# -------------------------------------------------------------- reductions
+@pytest.mark.parametrize("n", [1, 2, 1023, 65536, (1 << 22) + 9])
+def test_reduce_f64(dja, n):
+    x = philox.fill_uniform_f64(n, seed=31)
+    d = dja.distribute(x)
+    chunks = [x]
+    assert abs(dja.dsum(d) - oops.oracle_reduce("identity", "add", chunks)) \
+        <= 1e-12 * max(1.0, abs(x.sum()))
+    assert dja.dmaximum(d) == x.max()
+    assert dja.dminimum(d) == x.min()
+    assert dja.dextrema(d) == (x.min(), x.max())
+    s2 = dja.mapreduce("abs2", "add", d)
+    assert abs(s2 - (x * x).sum()) <= 1e-12 * (x * x).sum()
+    d.close()
+
+
+def test_reduce_prod(dja):
+    x = philox.fill_uniform_f64(1000, seed=32) * 0.04 + 0.98
+    d = dja.distribute(x)
+    assert abs(dja.dprod(d) - x.prod()) <= 1e-12 * abs(x.prod())
+    d.close()
+
+
+def test_reduce_i64_exact(dja):
+    with np.errstate(over="ignore"):
+        x = philox.fill_int64((1 << 20) + 7, seed=33)
+        d = dja.distribute(x)
+        # wrap-exact sum (test/darray.jl:286-294 exactness contract)
+        assert dja.dsum(d) == int(x.sum())
+        assert dja.dmaximum(d) == int(x.max())
+        assert dja.dminimum(d) == int(x.min())
+        d.close()
+
+
+def test_reduce_f32(dja):
+    x = philox.fill_uniform_f32(1 << 20, seed=34)
+    d = dja.distribute(x)
+    ref = (x.astype(np.float64) ** 2).sum()
+    got = dja.mapreduce("abs2", "add", d)
+    assert abs(got - ref) / ref < 1e-5
+    d.close()
+
+
+def test_reduce_empty(dja):
+    d = dja.dzeros((0,))
+    assert dja.dsum(d) == 0.0
+    assert dja.dprod(d) == 1.0
+    with pytest.raises(Exception):
+        dja.dmaximum(d)
+    d.close()
+
+
+def test_mean(dja):
+    x = philox.fill_uniform_f64(10000, seed=35)
+    d = dja.distribute(x)
+    assert abs(dja.dmean(d) - x.mean()) < 1e-13
+    d.close()
+
+
+def test_dot_norm(dja):
+    x = philox.fill_uniform_f64(20000, seed=36)
+    y = philox.fill_uniform_f64(20000, seed=37)
+    dx, dy = dja.distribute(x), dja.distribute(y)
+    assert abs(dja.ddot(dx, dy) - x @ y) <= 1e-12 * abs(x @ y)
+    assert abs(dja.dnorm(dx) - np.linalg.norm(x)) < 1e-10
+    assert abs(dja.dnorm(dx, 1) - np.abs(x).sum()) < 1e-9
+    assert dja.dnorm(dx, float("inf")) == np.abs(x).max()
+    dx.close(); dy.close()
+
+
+# ------------------------------------------------------------------- gemm
+def _gemm_case(dja, m, k, n, alpha=1.0, rtol=1e-12):
+    A = np.asfortranarray(philox.fill_uniform_f64(m * k, 41)
+                          .reshape(m, k, order="F"))
+    B = np.asfortranarray(philox.fill_uniform_f64(k * n, 42)
+                          .reshape(k, n, order="F"))
+    dA, dB = dja.distribute(A), dja.distribute(B)
+    C = dja.dmatmul(dA, dB, alpha=alpha)
+    got = C.localpart()
+    ref = alpha * (A @ B)
+    err = np.abs(got - ref).max() / np.abs(ref).max()
+    assert err < rtol, "gemm %dx%dx%d rel err %g" % (m, k, n, err)
+    C.close(); dA.close(); dB.close()
+
+
+def test_gemm_naive_path(dja):
+    _gemm_case(dja, 60, 50, 40)
+    _gemm_case(dja, 1, 7, 3)
+    _gemm_case(dja, 130, 33, 65)
+
+
+def test_gemm_mfma_path(dja):
+    _gemm_case(dja, 256, 256, 256)
+    _gemm_case(dja, 384, 128, 256)
+    _gemm_case(dja, 128, 1024, 128, alpha=2.0)
+
+
+def test_gemm_beta_via_abi(dja):
+    import distributedarrays_jl_amd as _dja
+    from distributedarrays_jl_amd._ffi import lib, check
+    m = k = n = 128
+    A = np.asfortranarray(philox.fill_uniform_f64(m * k, 43)
+                          .reshape(m, k, order="F"))
+    B = np.asfortranarray(philox.fill_uniform_f64(k * n, 44)
+                          .reshape(k, n, order="F"))
+    C0 = np.asfortranarray(philox.fill_uniform_f64(m * n, 45)
+                           .reshape(m, n, order="F"))
+    dA, dB = _dja.distribute(A), _dja.distribute(B)
+    dC = _dja.distribute(C0)
+    check(lib.da_gemm_f64(dC._ptr(), dA._ptr(), dB._ptr(), m, n, k,
+                          m, k, m, 2.0, 0.5))
+    check(lib.da_synchronize())
+    ref = 2.0 * (A @ B) + 0.5 * C0
+    got = dC.localpart()
+    assert np.abs(got - ref).max() / np.abs(ref).max() < 1e-12
+    dA.close(); dB.close(); dC.close()
+
+
+# ------------------------------------------------------ DArray round trips
+def test_distribute_collect_roundtrip(dja):
+    a = philox.fill_uniform_f64(60 * 77, 51).reshape(60, 77, order="F")
+    d = dja.distribute(a)
+    assert np.array_equal(d.collect(), a)
+    assert d == dja.distribute(a)
+    d.close()
+
+
+def test_leak_check(dja):
+    import distributedarrays_jl_amd as _dja
+    _dja.d_closeall()
+    base = _dja.bytes_in_use()
+    d = _dja.drand((4096,), "f64")
+    assert _dja.bytes_in_use() > base
+    d.close()
+    assert _dja.bytes_in_use() == base
+
+
+def test_events_api(dja):
+    from distributedarrays_jl_amd._ffi import lib, check
+    e0, e1 = ctypes.c_void_p(), ctypes.c_void_p()
+    check(lib.da_event_create(ctypes.byref(e0)))
+    check(lib.da_event_create(ctypes.byref(e1)))
+    check(lib.da_event_record(e0))
+    d = dja.drand((1 << 20,), "f64")
+    check(lib.da_event_record(e1))
+    ms = ctypes.c_float()
+    check(lib.da_event_elapsed(e0, e1, ctypes.byref(ms)))
+    assert ms.value >= 0.0
+    check(lib.da_event_destroy(e0))
+    check(lib.da_event_destroy(e1))
+    d.close()
