@@ -45,14 +45,15 @@ def main():
     if world != args.gpus and "WORLD_SIZE" in os.environ:
         args.gpus = world
 
+    # import order matters: our package pins /opt/rocm's HIP runtime
+    # before torch loads its bundled one (see _ffi.py)
+    import distributedarrays_jl_amd as dja
+    from distributedarrays_jl_amd._ffi import lib, check
     import torch
     import torch.distributed as td
     if world > 1:
         os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
         td.init_process_group("gloo", rank=rank, world_size=world)
-
-    import distributedarrays_jl_amd as dja
-    from distributedarrays_jl_amd._ffi import lib, check
     dja.comm.init()
 
     def barrier():

@@ -17,6 +17,16 @@ if not os.path.exists(_SO):
         "or `make -C distributedarrays_jl_amd/csrc`. There is no CPU "
         "fallback: the HIP extension IS the compute path." % _SO)
 
+# Load the system ROCm HIP runtime FIRST: torch wheels bundle their own
+# libamdhip64.so.7 (rocm7.0) and whichever object owns that soname first
+# wins the process; binding our gfx950 library against torch's bundled
+# runtime makes device enumeration fail.  Importing this package before
+# torch pins the /opt/rocm runtime (bench.py and tests do so).
+try:
+    ctypes.CDLL("/opt/rocm/lib/libamdhip64.so", mode=ctypes.RTLD_GLOBAL)
+except OSError:
+    pass
+
 lib = ctypes.CDLL(_SO)
 
 u64 = ctypes.c_uint64

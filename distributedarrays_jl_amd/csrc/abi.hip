@@ -281,6 +281,13 @@ int da_gemm_f64(void* C, const void* A, const void* B,
                            st().stream);
 }
 
+/* debug-only: MFMA lane-map probe (not part of the public ABI) */
+int dbg_mfma_probe_f64(const void* A, const void* B, void* out_c,
+                       void* out_raw) {
+    DA_REQUIRE_INIT();
+    return dbg_mfma_probe_impl(A, B, out_c, out_raw, st().stream);
+}
+
 /* ---- point-to-point --------------------------------------------------- */
 
 int da_group_start(void) {

@@ -139,6 +139,39 @@ __global__ void scale_c(double* __restrict__ C, int64_t m, int64_t n,
     C[j * ldc + i] = (beta == 0.0) ? 0.0 : beta * C[j * ldc + i];
 }
 
+// Debug probe: one v_mfma_f64_16x16x4_f64 with the lane mapping this
+// file assumes.  out_c = 16x16 col-major under the assumed C map;
+// out_raw = acc[q] per (lane,q) so a test can reverse-engineer the true
+// map if the assumption is wrong.  Exported as dbg_mfma_probe_f64.
+__global__ void mfma_probe_kernel(const double* __restrict__ A,
+                                  const double* __restrict__ B,
+                                  double* __restrict__ out_c,
+                                  double* __restrict__ out_raw) {
+    int l = threadIdx.x;
+    double a = A[(l & 15) + 16 * (l >> 4)];   // A[row, k] col-major 16x4
+    double b = B[(l >> 4) + 4 * (l & 15)];    // B[k, col] col-major 4x16
+    f64x4 acc = {0.0, 0.0, 0.0, 0.0};
+    acc = __builtin_amdgcn_mfma_f64_16x16x4f64(a, b, acc, 0, 0, 0);
+#pragma unroll
+    for (int q = 0; q < 4; ++q) {
+        out_raw[l * 4 + q] = acc[q];
+        int row = (l >> 4) * 4 + q, col = l & 15;
+        out_c[row + 16 * col] = acc[q];
+    }
+}
+
+extern "C" int dbg_mfma_probe_f64(const void* A, const void* B,
+                                  void* out_c, void* out_raw);
+int dbg_mfma_probe_impl(const void* A, const void* B, void* out_c,
+                        void* out_raw, hipStream_t s) {
+    hipLaunchKernelGGL(mfma_probe_kernel, dim3(1), dim3(64), 0, s,
+                       (const double*)A, (const double*)B,
+                       (double*)out_c, (double*)out_raw);
+    DA_CHECK_HIP(hipGetLastError());
+    DA_CHECK_HIP(hipStreamSynchronize(s));
+    return 0;
+}
+
 int launch_gemm_f64(void* Cv, const void* Av, const void* Bv,
                     int64_t m, int64_t n, int64_t k,
                     int64_t lda, int64_t ldb, int64_t ldc,

@@ -12,6 +12,12 @@ def pytest_configure(config):
 
 def have_gpu():
     try:
+        # load our HIP library (and /opt/rocm's runtime) BEFORE torch —
+        # torch's bundled rocm7.0 runtime must not own the hip soname
+        import distributedarrays_jl_amd._ffi  # noqa: F401
+    except Exception:
+        pass
+    try:
         import torch
         return torch.cuda.is_available()
     except Exception:
