@@ -27,11 +27,24 @@ def test_mfma_f64_lane_map():
     fn.argtypes = [ctypes.c_void_p] * 4
     fn.restype = ctypes.c_int
 
-    def p(a):
+    def hp(a):
         return a.ctypes.data_as(ctypes.c_void_p)
 
-    rc = fn(p(A), p(B), p(outc), p(outraw))
+    def dev(nbytes):
+        pp = ctypes.c_void_p()
+        assert lib.da_alloc(nbytes, 0, ctypes.byref(pp)) == 0
+        return pp
+
+    dA, dB = dev(16 * 4 * 8), dev(4 * 16 * 8)
+    dC, dR = dev(16 * 16 * 8), dev(64 * 4 * 8)
+    assert lib.da_h2d(dA, hp(A), 16 * 4 * 8) == 0
+    assert lib.da_h2d(dB, hp(B), 4 * 16 * 8) == 0
+    rc = fn(dA, dB, dC, dR)
     assert rc == 0, rc
+    assert lib.da_d2h(dC, hp(outc), 16 * 16 * 8) == 0
+    assert lib.da_d2h(dR, hp(outraw), 64 * 4 * 8) == 0
+    for d in (dA, dB, dC, dR):
+        lib.da_free(d)
 
     if np.allclose(outc, ref, rtol=1e-13):
         return  # assumed mapping is correct
