@@ -95,19 +95,19 @@ void gemm_f64_mfma(const double* __restrict__ A, const double* __restrict__ B,
         __syncthreads();
     }
 
-    // Epilogue: C/D fragment map col = lane&15, row = (lane>>4)*4 + j
-    // (shape-determined layout, dtype-independent on gfx950).
+    // Epilogue: v_mfma_f64_16x16x4_f64 C/D fragment map (probed on
+    // hardware, tests/test_gpu_mfma_probe.py):
+    //   col = lane & 15, row = 4*q + (lane >> 4)   (q = acc register)
 #pragma unroll
     for (int i = 0; i < 4; ++i) {
 #pragma unroll
         for (int j = 0; j < 4; ++j) {
             int64_t col = bn + wc + j * 16 + l16;
-            int64_t row0 = bm + wr + i * 16 + l4 * 4;
-            double* cp = C + col * ldc + row0;
+            double* cp = C + col * ldc + bm + wr + i * 16 + l4;
 #pragma unroll
             for (int q = 0; q < 4; ++q) {
                 double v = alpha * acc[i][j][q];
-                cp[q] = (beta == 0.0) ? v : v + beta * cp[q];
+                cp[4 * q] = (beta == 0.0) ? v : v + beta * cp[4 * q];
             }
         }
     }
@@ -155,7 +155,7 @@ __global__ void mfma_probe_kernel(const double* __restrict__ A,
 #pragma unroll
     for (int q = 0; q < 4; ++q) {
         out_raw[l * 4 + q] = acc[q];
-        int row = (l >> 4) * 4 + q, col = l & 15;
+        int row = 4 * q + (l >> 4), col = l & 15;   // probed f64 C/D map
         out_c[row + 16 * col] = acc[q];
     }
 }
