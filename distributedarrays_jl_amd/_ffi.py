@@ -42,6 +42,8 @@ _sigs = {
     "da_nranks": ([], i32),
     "da_alloc": ([u64, i32, ctypes.POINTER(ptr)], i32),
     "da_free": ([ptr], i32),
+    "da_pool_trim": ([], i32),
+    "da_pool_bytes": ([], u64),
     "da_h2d": ([ptr, ptr, u64], i32),
     "da_d2h": ([ptr, ptr, u64], i32),
     "da_d2d": ([ptr, ptr, u64], i32),

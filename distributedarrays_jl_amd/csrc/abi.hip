@@ -66,6 +66,9 @@ using namespace da;
 
 extern "C" {
 
+static uint64_t round_sz(uint64_t nbytes);
+static int pool_trim_locked();
+
 /* ---- lifecycle ------------------------------------------------------- */
 
 int da_init(int device, int rank, int nranks, const char* rccl_uid_path) {
@@ -119,9 +122,10 @@ int da_shutdown(void) {
     st().scratch_bytes = st().partials_bytes = 0;
     {
         std::lock_guard<std::mutex> g(st().mem_mtx);
-        for (auto& kv : st().allocs) hipFree(kv.first);
+        for (auto& kv : st().allocs) (void)hipFree(kv.first);
         st().allocs.clear();
         st().bytes_in_use = 0;
+        (void)pool_trim_locked();
     }
     if (st().stream) { hipStreamDestroy(st().stream); st().stream = nullptr; }
     st().inited = false;
@@ -133,17 +137,53 @@ int da_nranks(void) { return st().nranks; }
 
 /* ---- memory ----------------------------------------------------------- */
 
+static uint64_t round_sz(uint64_t nbytes) {
+    if (nbytes == 0) nbytes = 1;   // empty chunks keep a real handle
+    return (nbytes + 255) & ~(uint64_t)255;
+}
+
+static int pool_trim_locked() {
+    for (auto& kv : st().pool)
+        for (void* p : kv.second)
+            if (hipFree(p) != hipSuccess)
+                return set_err(-5, "da_pool_trim: hipFree failed");
+    st().pool.clear();
+    st().pool_bytes = 0;
+    return 0;
+}
+
 int da_alloc(uint64_t nbytes, int dtype, void** chunk) {
     DA_REQUIRE_INIT();
     if (!chunk) return set_err(-3, "da_alloc: null out");
     (void)dtype;
+    uint64_t sz = round_sz(nbytes);
     void* p = nullptr;
-    if (nbytes == 0) nbytes = 1;   // empty chunks keep a real handle
-    DA_CHECK_HIP(hipMalloc(&p, nbytes));
     {
         std::lock_guard<std::mutex> g(st().mem_mtx);
-        st().allocs[p] = nbytes;
-        st().bytes_in_use += nbytes;
+        auto it = st().pool.find(sz);
+        if (it != st().pool.end() && !it->second.empty()) {
+            p = it->second.back();
+            it->second.pop_back();
+            st().pool_bytes -= sz;
+        }
+    }
+    if (!p) {
+        hipError_t e = hipMalloc(&p, sz);
+        if (e != hipSuccess) {   // trim the cache and retry once
+            std::lock_guard<std::mutex> g(st().mem_mtx);
+            int rc = pool_trim_locked();
+            if (rc) return rc;
+            e = hipMalloc(&p, sz);
+            if (e != hipSuccess)
+                return set_err(-(1000 + (int)e), "da_alloc(%llu): %s",
+                               (unsigned long long)sz,
+                               hipGetErrorString(e));
+        }
+    }
+    {
+        std::lock_guard<std::mutex> g(st().mem_mtx);
+        st().allocs[p] = sz;
+        st().bytes_in_use += sz;
     }
     *chunk = p;
     return 0;
@@ -152,16 +192,28 @@ int da_alloc(uint64_t nbytes, int dtype, void** chunk) {
 int da_free(void* chunk) {
     DA_REQUIRE_INIT();
     if (!chunk) return 0;
-    {
-        std::lock_guard<std::mutex> g(st().mem_mtx);
-        auto it = st().allocs.find(chunk);
-        if (it == st().allocs.end())
-            return set_err(-3, "da_free: unknown chunk %p", chunk);
-        st().bytes_in_use -= it->second;
-        st().allocs.erase(it);
-    }
-    DA_CHECK_HIP(hipFree(chunk));
+    std::lock_guard<std::mutex> g(st().mem_mtx);
+    auto it = st().allocs.find(chunk);
+    if (it == st().allocs.end())
+        return set_err(-3, "da_free: unknown chunk %p", chunk);
+    uint64_t sz = it->second;
+    st().bytes_in_use -= sz;
+    st().allocs.erase(it);
+    st().pool[sz].push_back(chunk);   // cache; release via da_pool_trim
+    st().pool_bytes += sz;
     return 0;
+}
+
+int da_pool_trim(void) {
+    DA_REQUIRE_INIT();
+    DA_CHECK_HIP(hipStreamSynchronize(st().stream));
+    std::lock_guard<std::mutex> g(st().mem_mtx);
+    return pool_trim_locked();
+}
+
+uint64_t da_pool_bytes(void) {
+    std::lock_guard<std::mutex> g(st().mem_mtx);
+    return st().pool_bytes;
 }
 
 uint64_t da_bytes_in_use(void) {

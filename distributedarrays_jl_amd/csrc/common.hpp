@@ -31,6 +31,11 @@ struct State {
     std::mutex mem_mtx;
     std::unordered_map<void*, uint64_t> allocs;
     uint64_t bytes_in_use = 0;
+    // caching allocator: freed chunks kept per exact (256 B-rounded) size
+    // — hipMalloc/hipFree of multi-GiB staging buffers costs ~100 ms,
+    // which dominated the matmul leg (profiles/r01_kernel_stats.md)
+    std::unordered_map<uint64_t, std::vector<void*>> pool;
+    uint64_t pool_bytes = 0;
 };
 
 State& st();

@@ -1,4 +1,5 @@
 // kernels_gemm.hip — local dense C = alpha*A*B + beta*C, Float64,
+#include <stdlib.h>
 // COLUMN-major (Julia layout), MFMA-tiled for gfx950 (CDNA4).
 //
 // Replaces the worker-side `localpart(A) * Bjk` of
@@ -113,6 +114,127 @@ void gemm_f64_mfma(const double* __restrict__ A, const double* __restrict__ B,
     }
 }
 
+// v2: same 128x128x16 tiling with (a) a register-prefetch pipeline —
+// tile t+1's global loads issue before tile t's MFMA phase and land in
+// LDS after the barrier (write-after-barrier form of guideline T14) —
+// and (b) a bijective XCD-aware block remap for per-XCD L2 locality
+// (cdna_hip_programming.md §5: XCD swizzle, +10% when HBM-bound).
+__global__ __launch_bounds__(256, 2)
+void gemm_f64_mfma_v2(const double* __restrict__ A,
+                      const double* __restrict__ B,
+                      double* __restrict__ C, int64_t m, int64_t n,
+                      int64_t k, int64_t lda, int64_t ldb, int64_t ldc,
+                      double alpha, double beta) {
+    __shared__ double As[BM * LSTR];
+    __shared__ double Bs[BN * LSTR];
+
+    const int tid = threadIdx.x;
+    const int wave = tid >> 6;
+    const int lane = tid & 63;
+    const int l16 = lane & 15;
+    const int l4 = lane >> 4;
+    const int wr = (wave >> 1) * 64;
+    const int wc = (wave & 1) * 64;
+
+    // bijective XCD remap (dispatcher places block b on XCD b%8)
+    const int gx = gridDim.x, nwg = gridDim.x * gridDim.y;
+    int w = blockIdx.y * gx + blockIdx.x;
+    int q = nwg >> 3, rmd = nwg & 7, xcd = w & 7, idx = w >> 3;
+    int sw = (xcd < rmd ? xcd * (q + 1) : rmd * (q + 1) + (xcd - rmd) * q)
+             + idx;
+    const int64_t bm = (int64_t)(sw % gx) * BM;
+    const int64_t bn = (int64_t)(sw / gx) * BN;
+
+    f64x4 acc[4][4];
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+#pragma unroll
+        for (int j = 0; j < 4; ++j) acc[i][j] = {0.0, 0.0, 0.0, 0.0};
+
+    // per-thread staging coordinates (4 pieces of 2 doubles each side)
+    int a_c[4], a_r2[4], b_n[4], b_k2[4];
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+        int idx2 = tid + r * 256;
+        a_c[r] = idx2 >> 6;
+        a_r2[r] = (idx2 & 63) * 2;
+        b_n[r] = idx2 >> 3;
+        b_k2[r] = (idx2 & 7) * 2;
+    }
+
+    f64x2 pa[4], pb[4];
+    const int64_t ktiles = k / BK;
+    // prologue: tile 0 -> regs -> LDS
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+        pa[r] = *reinterpret_cast<const f64x2*>(A + (int64_t)a_c[r] * lda
+                                                + bm + a_r2[r]);
+        pb[r] = *reinterpret_cast<const f64x2*>(B + (bn + b_n[r]) * ldb
+                                                + b_k2[r]);
+    }
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+        As[(a_r2[r] + 0) * LSTR + a_c[r]] = pa[r].x;
+        As[(a_r2[r] + 1) * LSTR + a_c[r]] = pa[r].y;
+        *reinterpret_cast<f64x2*>(&Bs[b_n[r] * LSTR + b_k2[r]]) = pb[r];
+    }
+
+    for (int64_t kt = 0; kt < ktiles; ++kt) {
+        __syncthreads();   // LDS tile kt visible to all
+        if (kt + 1 < ktiles) {
+            const int64_t k0 = (kt + 1) * BK;
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                pa[r] = *reinterpret_cast<const f64x2*>(
+                    A + (k0 + a_c[r]) * lda + bm + a_r2[r]);
+                pb[r] = *reinterpret_cast<const f64x2*>(
+                    B + (bn + b_n[r]) * ldb + k0 + b_k2[r]);
+            }
+        }
+#pragma unroll
+        for (int kk = 0; kk < 4; ++kk) {
+            double a[4], b[4];
+            const int kof = kk * 4 + l4;
+#pragma unroll
+            for (int i = 0; i < 4; ++i)
+                a[i] = As[(wr + i * 16 + l16) * LSTR + kof];
+#pragma unroll
+            for (int j = 0; j < 4; ++j)
+                b[j] = Bs[(wc + j * 16 + l16) * LSTR + kof];
+#pragma unroll
+            for (int i = 0; i < 4; ++i)
+#pragma unroll
+                for (int j = 0; j < 4; ++j)
+                    acc[i][j] = __builtin_amdgcn_mfma_f64_16x16x4f64(
+                        a[i], b[j], acc[i][j], 0, 0, 0);
+        }
+        __syncthreads();   // MFMA phase done; LDS reusable
+        if (kt + 1 < ktiles) {
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                As[(a_r2[r] + 0) * LSTR + a_c[r]] = pa[r].x;
+                As[(a_r2[r] + 1) * LSTR + a_c[r]] = pa[r].y;
+                *reinterpret_cast<f64x2*>(&Bs[b_n[r] * LSTR + b_k2[r]])
+                    = pb[r];
+            }
+        }
+    }
+
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+            int64_t col = bn + wc + j * 16 + l16;
+            double* cp = C + col * ldc + bm + wr + i * 16 + l4;
+#pragma unroll
+            for (int qq = 0; qq < 4; ++qq) {
+                double v = alpha * acc[i][j][qq];
+                cp[4 * qq] = (beta == 0.0) ? v : v + beta * cp[4 * qq];
+            }
+        }
+    }
+}
+
 // Naive fallback for arbitrary shapes (small parity chunks).
 __global__ void gemm_f64_naive(const double* __restrict__ A,
                                const double* __restrict__ B,
@@ -190,8 +312,17 @@ int launch_gemm_f64(void* Cv, const void* Av, const void* Bv,
     }
     if (m % BM == 0 && n % BN == 0 && k % BK == 0) {
         dim3 g(m / BM, n / BN);
-        hipLaunchKernelGGL(gemm_f64_mfma, g, dim3(256), 0, s,
-                           A, B, C, m, n, k, lda, ldb, ldc, alpha, beta);
+        static int variant = -1;
+        if (variant < 0) {
+            const char* v = getenv("DA_GEMM_V");
+            variant = v ? atoi(v) : 2;
+        }
+        if (variant == 1)
+            hipLaunchKernelGGL(gemm_f64_mfma, g, dim3(256), 0, s,
+                               A, B, C, m, n, k, lda, ldb, ldc, alpha, beta);
+        else
+            hipLaunchKernelGGL(gemm_f64_mfma_v2, g, dim3(256), 0, s,
+                               A, B, C, m, n, k, lda, ldb, ldc, alpha, beta);
     } else {
         dim3 t(64, 4), g((m + 63) / 64, (n + 3) / 4);
         hipLaunchKernelGGL(gemm_f64_naive, g, t, 0, s,

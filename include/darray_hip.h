@@ -71,8 +71,13 @@ int da_nranks(void);
 
 /* ---- chunk memory (localpart storage; src/darray.jl:76-118 init /
  *      src/core.jl:67 release_localpart) ------------------------------- */
+/* Pooled: da_free caches the chunk for exact-size reuse (hipMalloc of
+ * multi-GiB staging buffers costs ~100 ms); da_pool_trim releases the
+ * cache to the driver.  bytes_in_use counts only live user chunks. */
 int da_alloc(uint64_t nbytes, int dtype, void** chunk);
 int da_free(void* chunk);
+int da_pool_trim(void);
+uint64_t da_pool_bytes(void);
 int da_h2d(void* chunk, const void* host, uint64_t nbytes);  /* distribute, darray.jl:544-555 */
 int da_d2h(const void* chunk, void* host, uint64_t nbytes);  /* collect / makelocal */
 int da_d2d(void* dst, const void* src, uint64_t nbytes);
