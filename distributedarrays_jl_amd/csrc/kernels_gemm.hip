@@ -119,14 +119,17 @@ void gemm_f64_mfma(const double* __restrict__ A, const double* __restrict__ B,
 // LDS after the barrier (write-after-barrier form of guideline T14) —
 // and (b) a bijective XCD-aware block remap for per-XCD L2 locality
 // (cdna_hip_programming.md §5: XCD swizzle, +10% when HBM-bound).
+template <int TBK>
 __global__ __launch_bounds__(256, 2)
 void gemm_f64_mfma_v2(const double* __restrict__ A,
                       const double* __restrict__ B,
                       double* __restrict__ C, int64_t m, int64_t n,
                       int64_t k, int64_t lda, int64_t ldb, int64_t ldc,
                       double alpha, double beta) {
-    __shared__ double As[BM * LSTR];
-    __shared__ double Bs[BN * LSTR];
+    constexpr int TSTR = TBK + 2;      // LDS row stride (bank-spread)
+    constexpr int NP = TBK / 4;        // staging pieces per thread/side
+    __shared__ double As[BM * TSTR];
+    __shared__ double Bs[BN * TSTR];
 
     const int tid = threadIdx.x;
     const int wave = tid >> 6;
@@ -151,40 +154,40 @@ void gemm_f64_mfma_v2(const double* __restrict__ A,
 #pragma unroll
         for (int j = 0; j < 4; ++j) acc[i][j] = {0.0, 0.0, 0.0, 0.0};
 
-    // per-thread staging coordinates (4 pieces of 2 doubles each side)
-    int a_c[4], a_r2[4], b_n[4], b_k2[4];
+    // per-thread staging coordinates (NP pieces of 2 doubles each side)
+    int a_c[NP], a_r2[NP], b_n[NP], b_k2[NP];
 #pragma unroll
-    for (int r = 0; r < 4; ++r) {
+    for (int r = 0; r < NP; ++r) {
         int idx2 = tid + r * 256;
         a_c[r] = idx2 >> 6;
         a_r2[r] = (idx2 & 63) * 2;
-        b_n[r] = idx2 >> 3;
-        b_k2[r] = (idx2 & 7) * 2;
+        b_n[r] = idx2 / (TBK / 2);
+        b_k2[r] = (idx2 % (TBK / 2)) * 2;
     }
 
-    f64x2 pa[4], pb[4];
-    const int64_t ktiles = k / BK;
+    f64x2 pa[NP], pb[NP];
+    const int64_t ktiles = k / TBK;
     // prologue: tile 0 -> regs -> LDS
 #pragma unroll
-    for (int r = 0; r < 4; ++r) {
+    for (int r = 0; r < NP; ++r) {
         pa[r] = *reinterpret_cast<const f64x2*>(A + (int64_t)a_c[r] * lda
                                                 + bm + a_r2[r]);
         pb[r] = *reinterpret_cast<const f64x2*>(B + (bn + b_n[r]) * ldb
                                                 + b_k2[r]);
     }
 #pragma unroll
-    for (int r = 0; r < 4; ++r) {
-        As[(a_r2[r] + 0) * LSTR + a_c[r]] = pa[r].x;
-        As[(a_r2[r] + 1) * LSTR + a_c[r]] = pa[r].y;
-        *reinterpret_cast<f64x2*>(&Bs[b_n[r] * LSTR + b_k2[r]]) = pb[r];
+    for (int r = 0; r < NP; ++r) {
+        As[(a_r2[r] + 0) * TSTR + a_c[r]] = pa[r].x;
+        As[(a_r2[r] + 1) * TSTR + a_c[r]] = pa[r].y;
+        *reinterpret_cast<f64x2*>(&Bs[b_n[r] * TSTR + b_k2[r]]) = pb[r];
     }
 
     for (int64_t kt = 0; kt < ktiles; ++kt) {
         __syncthreads();   // LDS tile kt visible to all
         if (kt + 1 < ktiles) {
-            const int64_t k0 = (kt + 1) * BK;
+            const int64_t k0 = (kt + 1) * TBK;
 #pragma unroll
-            for (int r = 0; r < 4; ++r) {
+            for (int r = 0; r < NP; ++r) {
                 pa[r] = *reinterpret_cast<const f64x2*>(
                     A + (k0 + a_c[r]) * lda + bm + a_r2[r]);
                 pb[r] = *reinterpret_cast<const f64x2*>(
@@ -192,15 +195,15 @@ void gemm_f64_mfma_v2(const double* __restrict__ A,
             }
         }
 #pragma unroll
-        for (int kk = 0; kk < 4; ++kk) {
+        for (int kk = 0; kk < TBK / 4; ++kk) {
             double a[4], b[4];
             const int kof = kk * 4 + l4;
 #pragma unroll
             for (int i = 0; i < 4; ++i)
-                a[i] = As[(wr + i * 16 + l16) * LSTR + kof];
+                a[i] = As[(wr + i * 16 + l16) * TSTR + kof];
 #pragma unroll
             for (int j = 0; j < 4; ++j)
-                b[j] = Bs[(wc + j * 16 + l16) * LSTR + kof];
+                b[j] = Bs[(wc + j * 16 + l16) * TSTR + kof];
 #pragma unroll
             for (int i = 0; i < 4; ++i)
 #pragma unroll
@@ -211,10 +214,10 @@ void gemm_f64_mfma_v2(const double* __restrict__ A,
         __syncthreads();   // MFMA phase done; LDS reusable
         if (kt + 1 < ktiles) {
 #pragma unroll
-            for (int r = 0; r < 4; ++r) {
-                As[(a_r2[r] + 0) * LSTR + a_c[r]] = pa[r].x;
-                As[(a_r2[r] + 1) * LSTR + a_c[r]] = pa[r].y;
-                *reinterpret_cast<f64x2*>(&Bs[b_n[r] * LSTR + b_k2[r]])
+            for (int r = 0; r < NP; ++r) {
+                As[(a_r2[r] + 0) * TSTR + a_c[r]] = pa[r].x;
+                As[(a_r2[r] + 1) * TSTR + a_c[r]] = pa[r].y;
+                *reinterpret_cast<f64x2*>(&Bs[b_n[r] * TSTR + b_k2[r]])
                     = pb[r];
             }
         }
@@ -312,16 +315,21 @@ int launch_gemm_f64(void* Cv, const void* Av, const void* Bv,
     }
     if (m % BM == 0 && n % BN == 0 && k % BK == 0) {
         dim3 g(m / BM, n / BN);
-        static int variant = -1;
+        static int variant = -1, bk = -1;
         if (variant < 0) {
             const char* v = getenv("DA_GEMM_V");
             variant = v ? atoi(v) : 2;
+            const char* b = getenv("DA_GEMM_BK");
+            bk = b ? atoi(b) : 16;
         }
         if (variant == 1)
             hipLaunchKernelGGL(gemm_f64_mfma, g, dim3(256), 0, s,
                                A, B, C, m, n, k, lda, ldb, ldc, alpha, beta);
+        else if (bk == 32 && k % 32 == 0)
+            hipLaunchKernelGGL(gemm_f64_mfma_v2<32>, g, dim3(256), 0, s,
+                               A, B, C, m, n, k, lda, ldb, ldc, alpha, beta);
         else
-            hipLaunchKernelGGL(gemm_f64_mfma_v2, g, dim3(256), 0, s,
+            hipLaunchKernelGGL(gemm_f64_mfma_v2<16>, g, dim3(256), 0, s,
                                A, B, C, m, n, k, lda, ldb, ldc, alpha, beta);
     } else {
         dim3 t(64, 4), g((m + 63) / 64, (n + 3) / 4);
