@@ -238,6 +238,123 @@ void gemm_f64_mfma_v2(const double* __restrict__ A,
     }
 }
 
+// v3: 128x128x16 tile, 512 threads = 8 waves of 64x32 sub-tiles
+// (acc = 8 fragments = 64 VGPRs vs v2's 128): 4 waves/SIMD co-residency
+// so one block's stage phase hides under another's MFMA phase.
+__global__ __launch_bounds__(512, 4)
+void gemm_f64_mfma_v3(const double* __restrict__ A,
+                      const double* __restrict__ B,
+                      double* __restrict__ C, int64_t m, int64_t n,
+                      int64_t k, int64_t lda, int64_t ldb, int64_t ldc,
+                      double alpha, double beta) {
+    __shared__ double As[BM * LSTR];
+    __shared__ double Bs[BN * LSTR];
+
+    const int tid = threadIdx.x;
+    const int wave = tid >> 6;
+    const int lane = tid & 63;
+    const int l16 = lane & 15;
+    const int l4 = lane >> 4;
+    const int wr = (wave >> 2) * 64;   // 2 wave-rows of 64
+    const int wc = (wave & 3) * 32;    // 4 wave-cols of 32
+
+    const int gx = gridDim.x, nwg = gridDim.x * gridDim.y;
+    int w = blockIdx.y * gx + blockIdx.x;
+    int q = nwg >> 3, rmd = nwg & 7, xcd = w & 7, idx = w >> 3;
+    int sw = (xcd < rmd ? xcd * (q + 1) : rmd * (q + 1) + (xcd - rmd) * q)
+             + idx;
+    const int64_t bm = (int64_t)(sw % gx) * BM;
+    const int64_t bn = (int64_t)(sw / gx) * BN;
+
+    f64x4 acc[4][2];
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+#pragma unroll
+        for (int j = 0; j < 2; ++j) acc[i][j] = {0.0, 0.0, 0.0, 0.0};
+
+    // staging: A/B tiles are 2048 doubles each; 512 threads x 2 f64x2
+    int a_c[2], a_r2[2], b_n[2], b_k2[2];
+#pragma unroll
+    for (int r = 0; r < 2; ++r) {
+        int idx2 = tid + r * 512;
+        a_c[r] = idx2 >> 6;
+        a_r2[r] = (idx2 & 63) * 2;
+        b_n[r] = idx2 >> 3;
+        b_k2[r] = (idx2 & 7) * 2;
+    }
+
+    f64x2 pa[2], pb[2];
+    const int64_t ktiles = k / BK;
+#pragma unroll
+    for (int r = 0; r < 2; ++r) {
+        pa[r] = *reinterpret_cast<const f64x2*>(A + (int64_t)a_c[r] * lda
+                                                + bm + a_r2[r]);
+        pb[r] = *reinterpret_cast<const f64x2*>(B + (bn + b_n[r]) * ldb
+                                                + b_k2[r]);
+    }
+#pragma unroll
+    for (int r = 0; r < 2; ++r) {
+        As[(a_r2[r] + 0) * LSTR + a_c[r]] = pa[r].x;
+        As[(a_r2[r] + 1) * LSTR + a_c[r]] = pa[r].y;
+        *reinterpret_cast<f64x2*>(&Bs[b_n[r] * LSTR + b_k2[r]]) = pb[r];
+    }
+
+    for (int64_t kt = 0; kt < ktiles; ++kt) {
+        __syncthreads();
+        if (kt + 1 < ktiles) {
+            const int64_t k0 = (kt + 1) * BK;
+#pragma unroll
+            for (int r = 0; r < 2; ++r) {
+                pa[r] = *reinterpret_cast<const f64x2*>(
+                    A + (k0 + a_c[r]) * lda + bm + a_r2[r]);
+                pb[r] = *reinterpret_cast<const f64x2*>(
+                    B + (bn + b_n[r]) * ldb + k0 + b_k2[r]);
+            }
+        }
+#pragma unroll
+        for (int kk = 0; kk < 4; ++kk) {
+            double a[4], b[2];
+            const int kof = kk * 4 + l4;
+#pragma unroll
+            for (int i = 0; i < 4; ++i)
+                a[i] = As[(wr + i * 16 + l16) * LSTR + kof];
+#pragma unroll
+            for (int j = 0; j < 2; ++j)
+                b[j] = Bs[(wc + j * 16 + l16) * LSTR + kof];
+#pragma unroll
+            for (int i = 0; i < 4; ++i)
+#pragma unroll
+                for (int j = 0; j < 2; ++j)
+                    acc[i][j] = __builtin_amdgcn_mfma_f64_16x16x4f64(
+                        a[i], b[j], acc[i][j], 0, 0, 0);
+        }
+        __syncthreads();
+        if (kt + 1 < ktiles) {
+#pragma unroll
+            for (int r = 0; r < 2; ++r) {
+                As[(a_r2[r] + 0) * LSTR + a_c[r]] = pa[r].x;
+                As[(a_r2[r] + 1) * LSTR + a_c[r]] = pa[r].y;
+                *reinterpret_cast<f64x2*>(&Bs[b_n[r] * LSTR + b_k2[r]])
+                    = pb[r];
+            }
+        }
+    }
+
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+#pragma unroll
+        for (int j = 0; j < 2; ++j) {
+            int64_t col = bn + wc + j * 16 + l16;
+            double* cp = C + col * ldc + bm + wr + i * 16 + l4;
+#pragma unroll
+            for (int qq = 0; qq < 4; ++qq) {
+                double v = alpha * acc[i][j][qq];
+                cp[4 * qq] = (beta == 0.0) ? v : v + beta * cp[4 * qq];
+            }
+        }
+    }
+}
+
 // Naive fallback for arbitrary shapes (small parity chunks).
 __global__ void gemm_f64_naive(const double* __restrict__ A,
                                const double* __restrict__ B,
@@ -324,6 +441,9 @@ int launch_gemm_f64(void* Cv, const void* Av, const void* Bv,
         }
         if (variant == 1)
             hipLaunchKernelGGL(gemm_f64_mfma, g, dim3(256), 0, s,
+                               A, B, C, m, n, k, lda, ldb, ldc, alpha, beta);
+        else if (variant == 3)
+            hipLaunchKernelGGL(gemm_f64_mfma_v3, g, dim3(512), 0, s,
                                A, B, C, m, n, k, lda, ldb, ldc, alpha, beta);
         else if (bk == 32 && k % 32 == 0)
             hipLaunchKernelGGL(gemm_f64_mfma_v2<32>, g, dim3(256), 0, s,
