@@ -22,7 +22,9 @@ from .darray import (DArray, dzeros, dones, dfill, drand, drandn,
                      bytes_in_use)
 from .ops import (map_, dmap, map2_, elementwise, broadcast_fma, axpy_,
                   add_, scale_, mapreduce, dsum, dprod, dmaximum, dminimum,
-                  dextrema, dmean, ddot, dnorm, dmatmul)
+                  dextrema, dmean, ddot, dnorm, dmatmul, dreduce_dims,
+                  dsum_dims, dprod_dims, dmaximum_dims, dminimum_dims,
+                  dmean_dims, dmatvec)
 
 __all__ = [
     "DArray", "DArrayError", "comm", "geometry", "plan",
@@ -31,4 +33,6 @@ __all__ = [
     "map_", "dmap", "map2_", "elementwise", "broadcast_fma", "axpy_",
     "add_", "scale_", "mapreduce", "dsum", "dprod", "dmaximum",
     "dminimum", "dextrema", "dmean", "ddot", "dnorm", "dmatmul",
+    "dreduce_dims", "dsum_dims", "dprod_dims", "dmaximum_dims",
+    "dminimum_dims", "dmean_dims", "dmatvec",
 ]

@@ -304,6 +304,13 @@ int da_reduce(int mapop, int redop, const void* src, uint64_t n, int dtype,
     return launch_reduce(mapop, redop, src, n, dtype, out, st().stream);
 }
 
+int da_reduce_dims(int mapop, int redop, const void* src, uint64_t inner,
+                   uint64_t axis, uint64_t outer, int dtype, void* dst) {
+    DA_REQUIRE_INIT();
+    return launch_reduce_dims(mapop, redop, src, inner, axis, outer,
+                              dtype, dst, st().stream);
+}
+
 int da_allreduce(void* inout, int count, int dtype, int redop) {
     DA_REQUIRE_INIT();
     if (count <= 0) return set_err(-3, "da_allreduce: bad count");

@@ -92,6 +92,9 @@ int launch_add(void* dest, const void* src, double scale, uint64_t n,
 int launch_scale(void* a, double s_, uint64_t n, int dtype, hipStream_t s);
 int launch_reduce(int mapop, int redop, const void* src, uint64_t n,
                   int dtype, void* out_host, hipStream_t s);
+int launch_reduce_dims(int mapop, int redop, const void* src,
+                       uint64_t inner, uint64_t axis, uint64_t outer,
+                       int dtype, void* dst, hipStream_t s);
 int launch_gemm_f64(void* C, const void* A, const void* B,
                     int64_t m, int64_t n, int64_t k,
                     int64_t lda, int64_t ldb, int64_t ldc,

@@ -154,6 +154,92 @@ static int do_reduce(int mapop, int redop, const T* src, uint64_t n,
     return 0;
 }
 
+// ---- dims-reduction (mapreduce.jl:42-94: per-chunk mapreduce(dims=..))
+// The chunk is viewed column-major as (inner, axis, outer):
+// src[i + a*inner + o*inner*axis] -> dst[i + o*inner], reduced over a.
+// Variant A (inner >= 64): one thread per (i,o), serial over axis —
+// coalesced across inner.  Variant B (inner < 64): one 64-lane wave per
+// (i,o), lanes stride the axis (coalesced along the axis), then a
+// wavefront shuffle tree.
+template <typename T>
+__global__ void reduce_dims_threads(int mapop, int redop,
+                                    const T* __restrict__ src,
+                                    uint64_t inner, uint64_t axis,
+                                    uint64_t outer, T* __restrict__ dst) {
+    uint64_t t = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    uint64_t total = inner * outer;
+    for (uint64_t e = t; e < total; e += stride) {
+        uint64_t i = e % inner, o = e / inner;
+        const T* p = src + i + o * inner * axis;
+        T acc = RedIdent<T>::get(redop);
+        for (uint64_t a = 0; a < axis; ++a)
+            acc = red_comb(redop, acc, mapf(mapop, p[a * inner]));
+        dst[i + o * inner] = acc;
+    }
+}
+
+template <typename T>
+__global__ void reduce_dims_waves(int mapop, int redop,
+                                  const T* __restrict__ src,
+                                  uint64_t inner, uint64_t axis,
+                                  uint64_t outer, T* __restrict__ dst) {
+    uint64_t wid = ((uint64_t)blockIdx.x * blockDim.x + threadIdx.x) / 64;
+    uint64_t nw = ((uint64_t)gridDim.x * blockDim.x) / 64;
+    int lane = threadIdx.x & 63;
+    uint64_t total = inner * outer;
+    for (uint64_t e = wid; e < total; e += nw) {
+        uint64_t i = e % inner, o = e / inner;
+        const T* p = src + i + o * inner * axis;
+        T acc = RedIdent<T>::get(redop);
+        for (uint64_t a = lane; a < axis; a += 64)
+            acc = red_comb(redop, acc, mapf(mapop, p[a * inner]));
+#pragma unroll
+        for (int off = 32; off > 0; off >>= 1)
+            acc = red_comb(redop, acc, (T)__shfl_down(acc, off, 64));
+        if (lane == 0) dst[i + o * inner] = acc;
+    }
+}
+
+template <typename T>
+static int do_reduce_dims(int mapop, int redop, const T* src,
+                          uint64_t inner, uint64_t axis, uint64_t outer,
+                          T* dst, hipStream_t s) {
+    uint64_t total = inner * outer;
+    if (total == 0) return 0;
+    // axis == 0 still runs: the loop body never executes and dst gets
+    // the fold identity.
+    if (inner >= 64) {
+        uint64_t want = (total + RTPB - 1) / RTPB;
+        int g = (int)(want < 1 ? 1 : (want > 8192 ? 8192 : want));
+        hipLaunchKernelGGL(reduce_dims_threads<T>, dim3(g), dim3(RTPB), 0,
+                           s, mapop, redop, src, inner, axis, outer, dst);
+    } else {
+        uint64_t want = (total * 64 + RTPB - 1) / RTPB;
+        int g = (int)(want < 1 ? 1 : (want > 8192 ? 8192 : want));
+        hipLaunchKernelGGL(reduce_dims_waves<T>, dim3(g), dim3(RTPB), 0,
+                           s, mapop, redop, src, inner, axis, outer, dst);
+    }
+    DA_CHECK_HIP(hipGetLastError());
+    return 0;
+}
+
+int launch_reduce_dims(int mapop, int redop, const void* src,
+                       uint64_t inner, uint64_t axis, uint64_t outer,
+                       int dtype, void* dst, hipStream_t s) {
+    if (mapop < 0 || mapop > DA_REDF_ABS2 || redop < 0 || redop > DA_RED_MAX)
+        return set_err(-3, "da_reduce_dims: bad op (%d,%d)", mapop, redop);
+    switch (dtype) {
+    case DA_F64: return do_reduce_dims<double>(mapop, redop,
+        (const double*)src, inner, axis, outer, (double*)dst, s);
+    case DA_F32: return do_reduce_dims<float>(mapop, redop,
+        (const float*)src, inner, axis, outer, (float*)dst, s);
+    case DA_I64: return do_reduce_dims<int64_t>(mapop, redop,
+        (const int64_t*)src, inner, axis, outer, (int64_t*)dst, s);
+    }
+    return set_err(-3, "da_reduce_dims: bad dtype %d", dtype);
+}
+
 int launch_reduce(int mapop, int redop, const void* src, uint64_t n,
                   int dtype, void* out_host, hipStream_t s) {
     if (mapop < 0 || mapop > DA_REDF_ABS2 || redop < 0 || redop > DA_RED_MAX)

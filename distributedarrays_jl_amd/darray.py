@@ -37,7 +37,7 @@ class DArray:
     """Block-distributed dense array; localpart is an HBM chunk."""
 
     def __init__(self, dims, dtype="f64", dist=None, init=None,
-                 _alloc=True):
+                 ranks=None, _alloc=True):
         _auto_init()
         rank, nr = comm.rank_info()
         dims = tuple(int(d) for d in dims)
@@ -57,11 +57,20 @@ class DArray:
         self.nranks = nr
         self.nchunks = np_chunks
         self.idxs, self.cuts = geometry.chunk_indices(dims, dist)
+        # chunk c is owned by ranks[c] (default identity) — mirrors the
+        # reference's pids array (e.g. reducedim_initarray stores the
+        # result on A.pids[region -> 1:1], mapreduce.jl:42-50)
+        if ranks is None:
+            ranks = list(range(np_chunks))
+        if len(ranks) != np_chunks:
+            raise ValueError("ranks must list one owner per chunk")
+        self.ranks = ranks
         self.id = _next_id[0]
         _next_id[0] += 1
         self._chunk = None
-        if rank < np_chunks:
-            self.lidx = self.idxs[rank]
+        self.lchunk = ranks.index(rank) if rank in ranks else None
+        if self.lchunk is not None:
+            self.lidx = self.idxs[self.lchunk]
             self.lshape = geometry.shape_of(self.lidx)
             self.lnumel = geometry.nelems(self.lidx)
         else:
@@ -186,7 +195,7 @@ class DArray:
         the reference's remotecall gather (darray.jl:574ff)."""
         npdt = np.dtype(NUMPY_DTYPES[self.dtype])
         local = self.localpart()
-        if self.nchunks == 1:
+        if self.nchunks == 1 and self.lchunk is not None:
             out = np.zeros(self.dims, dtype=npdt, order="F")
             if self.lnumel:
                 sl = tuple(slice(lo, hi) for lo, hi in self.lidx)
@@ -200,9 +209,9 @@ class DArray:
         gathered = [None] * self.nranks
         td.all_gather_object(gathered, local)
         out = np.zeros(self.dims, dtype=npdt, order="F")
-        for r in range(self.nchunks):
-            sl = tuple(slice(lo, hi) for lo, hi in self.idxs[r])
-            out[sl] = gathered[r]
+        for c in range(self.nchunks):
+            sl = tuple(slice(lo, hi) for lo, hi in self.idxs[c])
+            out[sl] = gathered[self.ranks[c]]
         return out
 
     def __eq__(self, other):

@@ -329,3 +329,223 @@ def dmatmul(A, B, alpha=1.0):
     if slab is not None:
         slab.free()
     return C
+
+
+# ------------------------------------------------------- dims-reductions
+_RED2MAP2 = {"add": "add", "mul": "mul", "min": "min2", "max": "max2"}
+
+
+def dreduce_dims(f, op, d, dims):
+    """mapreduce(f, op, D; dims=dims) — src/mapreduce.jl:42-94:
+    per-chunk kernel reduction (mapreducedim_within), then cross-chunk
+    combine onto the lowest-coordinate slab owners
+    (reducedim_initarray stores R on A.pids[region -> 1:1],
+    mapreducedim_between! pulls co-slabs and folds; here: grouped
+    ncclSend/Recv of partial slabs + ascending-source-rank elementwise
+    combine).  Result keeps reduced dims as size 1 (Julia keepdims)."""
+    if isinstance(dims, int):
+        dims = (dims,)
+    dims = tuple(sorted(set(int(a) for a in dims)))
+    nd = d.ndims
+    if any(a < 0 or a >= nd for a in dims):
+        raise DArrayError("dreduce_dims: bad dims %r" % (dims,))
+    if d.ranks != list(range(d.nchunks)):
+        raise DArrayError("dreduce_dims: source must have identity ranks")
+    dt = DTYPES[d.dtype]
+    esz = DTYPE_SIZE[d.dtype]
+
+    rdims = tuple(1 if a in dims else d.dims[a] for a in range(nd))
+    rdist = tuple(1 if a in dims else d.dist[a] for a in range(nd))
+    # owner of R chunk c = D-grid rank with reduced coords 0
+    nr_chunks = 1
+    for c in rdist:
+        nr_chunks *= c
+    owners = []
+    for lin in range(nr_chunks):
+        sub = list(geometry.grid_pos(lin, rdist))
+        owners.append(geometry.grid_rank(sub, d.dist))
+    R = DArray(rdims, d.dtype, rdist, ranks=owners)
+
+    me = d.rank
+    have_chunk = d.lchunk is not None and d.lnumel >= 0
+    partial = None
+    pshape = None
+    if d.lchunk is not None:
+        # sequential per-axis reduction; mapf only on the first pass
+        shape = list(d.lshape)
+        src_ptr = d._ptr()
+        cur = None
+        first = True
+        for a in dims:
+            inner = 1
+            for x in shape[:a]:
+                inner *= x
+            outer = 1
+            for x in shape[a + 1:]:
+                outer *= x
+            axis = shape[a]
+            nxt = _Buf(max(inner * outer, 1) * esz)
+            check(lib.da_reduce_dims(RED_FS[f if first else "identity"],
+                                     RED_OPS[op], src_ptr, inner, axis,
+                                     outer, dt, nxt.p))
+            if cur is not None:
+                cur.free()
+            cur = nxt
+            src_ptr = cur.p
+            shape[a] = 1
+            first = False
+        partial = cur
+        pshape = tuple(shape)
+
+    # exchange: group D-ranks by owner (ascending rank == ascending
+    # column-major reduced coords, the co-slab concatenation order of
+    # mapreducedim_between!, mapreduce.jl:71-81)
+    groups = {}
+    for src in range(d.nchunks):
+        sub = list(geometry.grid_pos(src, d.dist))
+        for a in dims:
+            sub[a] = 0
+        groups.setdefault(geometry.grid_rank(sub, d.dist), []).append(src)
+
+    my_owner_group = groups.get(me) if R.lchunk is not None else None
+    sends = []
+    recvs = {}
+    if d.lchunk is not None:
+        my_group_owner = None
+        sub = list(geometry.grid_pos(me, d.dist))
+        for a in dims:
+            sub[a] = 0
+        my_group_owner = geometry.grid_rank(sub, d.dist)
+        if my_group_owner != me:
+            sends.append((my_group_owner, partial))
+    if my_owner_group:
+        for src in my_owner_group:
+            if src != me:
+                recvs[src] = _Buf(max(R.lnumel, 1) * esz)
+    if sends or recvs:
+        check(lib.da_group_start())
+        for (dst, buf) in sends:
+            check(lib.da_send(buf.p, max(d2_numel(pshape), 1) * esz, dst))
+        for src, buf in recvs.items():
+            check(lib.da_recv(buf.p, max(R.lnumel, 1) * esz, src))
+        check(lib.da_group_end())
+
+    if my_owner_group:
+        # fold ascending source rank; owner (reduced coords 0) is first
+        first_src = my_owner_group[0]
+        if first_src == me:
+            if R.lnumel:
+                check(lib.da_d2d(R._ptr(), partial.p, R.lnumel * esz))
+        else:
+            if R.lnumel:
+                check(lib.da_d2d(R._ptr(), recvs[first_src].p,
+                                 R.lnumel * esz))
+        for src in my_owner_group[1:]:
+            buf = partial if src == me else recvs[src]
+            if R.lnumel:
+                check(lib.da_map2(MAP2_OP[_RED2MAP2[op]], R._ptr(),
+                                  R._ptr(), buf.p, R.lnumel, dt))
+    check(lib.da_synchronize())
+    if partial is not None:
+        partial.free()
+    for buf in recvs.values():
+        buf.free()
+    return R
+
+
+def d2_numel(shape):
+    n = 1
+    for x in shape:
+        n *= x
+    return n
+
+
+def dsum_dims(d, dims):
+    return dreduce_dims("identity", "add", d, dims)
+
+
+def dprod_dims(d, dims):
+    return dreduce_dims("identity", "mul", d, dims)
+
+
+def dmaximum_dims(d, dims):
+    return dreduce_dims("identity", "max", d, dims)
+
+
+def dminimum_dims(d, dims):
+    return dreduce_dims("identity", "min", d, dims)
+
+
+def dmean_dims(d, dims):
+    """mean(D; dims) rides the sum path (ext/StatisticsExt.jl:6)."""
+    if isinstance(dims, int):
+        dims = (dims,)
+    R = dsum_dims(d, dims)
+    nred = 1
+    for a in dims:
+        nred *= d.dims[a]
+    scale_(R, 1.0 / nred)
+    return R
+
+
+# ---------------------------------------------------------------- matvec
+def dmatvec(A, x, alpha=1.0):
+    """y = alpha * A * x — mul!(y, A, x) (linalg.jl:78-122): each rank
+    (i,j) multiplies its block by the x-slice of its column cut
+    (linalg.jl:91: xj shipped per tile), partial vectors travel to the
+    y owner (rank i = procs(A)[i,1]) and accumulate ascending j."""
+    import numpy as np
+    if A.dtype != "f64" or A.ndims != 2:
+        raise DArrayError("dmatvec: 2-D f64 only")
+    x = np.ascontiguousarray(np.asarray(x, dtype=np.float64))
+    if x.shape != (A.dims[1],):
+        raise DArrayError("dmatvec: x length %d != %d"
+                          % (x.shape[0], A.dims[1]))
+    m = A.dims[0]
+    I, J = A.dist
+    y = DArray((m,), "f64", (I,))
+    y.fill_(0.0)
+    r = A.rank
+    esz = 8
+    partial = None
+    if A.lchunk is not None and A.lnumel:
+        i, j = r % I, r // I
+        jlo, jhi = A.lidx[1]
+        xj = x[jlo:jhi]
+        xbuf = _Buf(max(xj.size, 1) * esz)
+        check(lib.da_h2d(xbuf.p, xj.ctypes.data_as(ctypes.c_void_p),
+                         xj.size * esz))
+        mloc, kloc = A.lshape
+        partial = _Buf(mloc * esz)
+        check(lib.da_gemm_f64(partial.p, A._ptr(), xbuf.p,
+                              mloc, 1, kloc, mloc, kloc, mloc, 1.0, 0.0))
+        xbuf.free()
+
+    sends, recvs = [], {}
+    if A.lchunk is not None and A.lnumel:
+        i, j = r % I, r // I
+        if j != 0:
+            sends.append((i, partial))
+    if r < I and A.nchunks > I:
+        for j in range(1, J):
+            src = r + I * j
+            recvs[src] = _Buf(max(y.lnumel, 1) * esz)
+    if sends or recvs:
+        check(lib.da_group_start())
+        for dst, buf in sends:
+            check(lib.da_send(buf.p, A.lshape[0] * esz, dst))
+        for src, buf in recvs.items():
+            check(lib.da_recv(buf.p, y.lnumel * esz, src))
+        check(lib.da_group_end())
+    if r < I and y.lnumel:
+        for j in range(J):
+            src = r + I * j
+            buf = partial if src == r else recvs[src]
+            check(lib.da_add(y._ptr(), buf.p, float(alpha), y.lnumel,
+                             DTYPES["f64"]))
+    check(lib.da_synchronize())
+    if partial is not None:
+        partial.free()
+    for buf in recvs.values():
+        buf.free()
+    return y

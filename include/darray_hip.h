@@ -113,6 +113,11 @@ int da_scale(void* a, double s, uint64_t n, int dtype);      /* rmul!, linalg.jl
  * n == 0 returns the fold identity. */
 int da_reduce(int mapop, int redop, const void* src, uint64_t n, int dtype,
               void* out);
+/* Per-chunk dims-reduction (src/mapreduce.jl:42-94, mapreducedim_within):
+ * chunk viewed column-major as (inner, axis, outer); dst gets
+ * inner*outer elements reduced over axis (identity when axis==0). */
+int da_reduce_dims(int mapop, int redop, const void* src, uint64_t inner,
+                   uint64_t axis, uint64_t outer, int dtype, void* dst);
 /* Cross-rank fold of scalar partials (the caller-side reduce(op, results)
  * of mapreduce.jl:34, re-expressed as an RCCL allreduce over xGMI).
  * inout is HOST memory of `count` dtype elements; nranks==1 is a no-op. */

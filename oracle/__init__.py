@@ -43,4 +43,4 @@ from .ops import (MAP_OPS, MAP2_OPS, RED_OPS, MAPRED_FS,
                   oracle_map, oracle_map2, oracle_chunk_reduce,
                   oracle_reduce, oracle_bcast_fma, oracle_axpy,
                   oracle_add, oracle_scale, oracle_matmul_blocked,
-                  make_chunks, assemble)
+                  oracle_reduce_dims, make_chunks, assemble)

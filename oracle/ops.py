@@ -170,6 +170,39 @@ def oracle_scale(a, s):
     return np.asfortranarray(a * np.asarray(s, a.dtype))
 
 
+def oracle_reduce_dims(mapop, redop, chunks, idxs, dims, red_axes,
+                       dtype=None):
+    """mapreduce(f, op, A; dims=red_axes) — mapreduce.jl:42-94:
+    per-chunk reduce with keepdims (mapreducedim_within), then combine
+    chunk partials along the reduced axes in ascending chunk-coordinate
+    order onto the lowest-coordinate slab (mapreducedim_between!).
+    Returns the full (keepdims) result array."""
+    import numpy as np
+    f = MAPRED_FS[mapop]
+    opf, init = RED_OPS[redop]
+    nd = len(dims)
+    rdims = tuple(1 if a in red_axes else dims[a] for a in range(nd))
+    dt = chunks[0].dtype if dtype is None else np.dtype(dtype)
+    out = np.full(rdims, init(dt), dtype=dt, order="F")
+    for ch, idx in zip(chunks, idxs):
+        if ch.size == 0:
+            continue
+        part = f(ch)
+        for a in sorted(red_axes):
+            if redop == "add":
+                part = part.sum(axis=a, keepdims=True, dtype=dt)
+            elif redop == "mul":
+                part = part.prod(axis=a, keepdims=True, dtype=dt)
+            elif redop == "min":
+                part = part.min(axis=a, keepdims=True)
+            else:
+                part = part.max(axis=a, keepdims=True)
+        sl = tuple(slice(0, 1) if a in red_axes else slice(lo, hi)
+                   for a, (lo, hi) in enumerate(idx))
+        out[sl] = opf(out[sl], part)
+    return np.asfortranarray(out)
+
+
 # ------------------------------------------------------------ chunk helpers
 def make_chunks(arr, idxs):
     """Slice a global array into chunks per chunk_idxs output."""

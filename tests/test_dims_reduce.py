@@ -1,0 +1,95 @@
+"""dims-reductions (SURVEY §8f row 1 / mapreduce.jl:42-94): oracle
+self-consistency on CPU, and GPU parity of the DArray-level op."""
+import numpy as np
+import pytest
+
+from oracle import philox, ops as oops, geometry as ogeo
+
+
+def test_oracle_reduce_dims_vs_numpy():
+    x = np.asfortranarray(philox.fill_uniform_f64(60 * 40, 1)
+                          .reshape(60, 40, order="F"))
+    for nr in (1, 2, 4, 8):
+        dist = ogeo.defaultdist_dims([60, 40], nr)
+        idxs, _ = ogeo.chunk_idxs([60, 40], dist)
+        chunks = oops.make_chunks(x, idxs)
+        for axes in [(0,), (1,), (0, 1)]:
+            got = oops.oracle_reduce_dims("identity", "add", chunks, idxs,
+                                          (60, 40), axes)
+            ref = x.sum(axis=axes, keepdims=True)
+            assert np.allclose(got, ref, rtol=1e-12), (nr, axes)
+            gmax = oops.oracle_reduce_dims("identity", "max", chunks, idxs,
+                                           (60, 40), axes)
+            assert np.array_equal(gmax, x.max(axis=axes, keepdims=True))
+
+
+def test_oracle_reduce_dims_int_exact():
+    with np.errstate(over="ignore"):
+        x = np.asfortranarray(philox.fill_int64(32 * 24, 2)
+                              .reshape(32, 24, order="F"))
+        dist = [2, 2]
+        idxs, _ = ogeo.chunk_idxs([32, 24], dist)
+        chunks = oops.make_chunks(x, idxs)
+        got = oops.oracle_reduce_dims("identity", "add", chunks, idxs,
+                                      (32, 24), (0,))
+        assert np.array_equal(got, x.sum(axis=0, keepdims=True))
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("shape,axes", [
+    ((128, 96), (0,)), ((128, 96), (1,)), ((128, 96), (0, 1)),
+    ((1000,), (0,)), ((17, 33), (0,)), ((64, 32, 16), (1,)),
+    ((64, 32, 16), (0, 2)),
+])
+def test_gpu_dims_reduce(shape, axes):
+    import distributedarrays_jl_amd as dja
+    dja.comm.init()
+    n = int(np.prod(shape))
+    x = np.asfortranarray(philox.fill_uniform_f64(n, 7)
+                          .reshape(shape, order="F"))
+    d = dja.distribute(x)
+    R = dja.dsum_dims(d, axes)
+    ref = x.sum(axis=axes, keepdims=True)
+    assert R.dims == ref.shape
+    got = R.collect()
+    assert np.allclose(got, ref, rtol=1e-12), (shape, axes)
+    R.close()
+    M = dja.dmaximum_dims(d, axes)
+    assert np.array_equal(M.collect(), x.max(axis=axes, keepdims=True))
+    M.close()
+    A = dja.dmean_dims(d, axes)
+    assert np.allclose(A.collect(), x.mean(axis=axes, keepdims=True),
+                       rtol=1e-12)
+    A.close()
+    d.close()
+
+
+@pytest.mark.gpu
+def test_gpu_dims_reduce_abs2():
+    import distributedarrays_jl_amd as dja
+    dja.comm.init()
+    x = np.asfortranarray(philox.fill_uniform_f64(96 * 64, 9)
+                          .reshape(96, 64, order="F"))
+    d = dja.distribute(x)
+    R = dja.dreduce_dims("abs2", "add", d, (0,))
+    assert np.allclose(R.collect(), (x * x).sum(axis=0, keepdims=True),
+                       rtol=1e-12)
+    R.close(); d.close()
+
+
+@pytest.mark.gpu
+def test_gpu_matvec():
+    import distributedarrays_jl_amd as dja
+    dja.comm.init()
+    m, k = 200, 150
+    A = np.asfortranarray(philox.fill_uniform_f64(m * k, 11)
+                          .reshape(m, k, order="F"))
+    x = philox.fill_uniform_f64(k, 12)
+    dA = dja.distribute(A)
+    y = dja.dmatvec(dA, x)
+    assert y.dims == (m,)
+    assert np.allclose(y.collect(), A @ x, rtol=1e-12)
+    y.close()
+    y2 = dja.dmatvec(dA, x, alpha=2.0)
+    assert np.allclose(y2.collect(), 2.0 * (A @ x), rtol=1e-12)
+    y2.close(); dA.close()
