@@ -16,11 +16,21 @@
 #include "common.hpp"
 #include "philox_device.hpp"
 #include <math.h>
+#include <stdlib.h>
 
 namespace da {
 
 constexpr int TPB = 256;
 constexpr int MAXBLOCKS = 8192;   // 1024 workgroups per XCD — fills the chip
+
+static inline bool use_nt() {
+    static int nt = -1;
+    if (nt < 0) {
+        const char* e = getenv("DA_NT");
+        nt = e ? atoi(e) : 0;
+    }
+    return nt == 1;
+}
 
 static inline int nblocks(uint64_t work) {
     uint64_t b = (work + TPB - 1) / TPB;
@@ -30,7 +40,9 @@ static inline int nblocks(uint64_t work) {
 }
 
 // ------------------------------------------------------------------- fill
-template <typename T>
+// NT = nontemporal stores on the write-only stream (gfx950 `nt` flag) —
+// A/B-gated via DA_NT; profiles/ record the measured winner.
+template <typename T, bool NT>
 __global__ void fill_kernel(T* __restrict__ p, T v, uint64_t n) {
     uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
     uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
@@ -39,23 +51,44 @@ __global__ void fill_kernel(T* __restrict__ p, T v, uint64_t n) {
     using V = T __attribute__((ext_vector_type(2)));
     V vv = {v, v};
     V* pv = reinterpret_cast<V*>(p);
-    for (uint64_t j = i; j < nv; j += stride) pv[j] = vv;
+    for (uint64_t j = i; j < nv; j += stride) {
+        if (NT) __builtin_nontemporal_store(vv, pv + j);
+        else pv[j] = vv;
+    }
     for (uint64_t j = 2 * nv + i; j < n; j += stride) p[j] = v;
 }
 
 int launch_fill(void* chunk, double v, uint64_t n, int dtype, hipStream_t s) {
     if (n == 0) return 0;
     int g = nblocks(n / 2 + 1);
+    if (use_nt()) {
+        switch (dtype) {
+        case DA_F64:
+            hipLaunchKernelGGL((fill_kernel<double, true>), dim3(g),
+                               dim3(TPB), 0, s, (double*)chunk, v, n); break;
+        case DA_F32:
+            hipLaunchKernelGGL((fill_kernel<float, true>), dim3(g),
+                               dim3(TPB), 0, s, (float*)chunk, (float)v, n);
+            break;
+        case DA_I64:
+            hipLaunchKernelGGL((fill_kernel<int64_t, true>), dim3(g),
+                               dim3(TPB), 0, s, (int64_t*)chunk,
+                               (int64_t)v, n); break;
+        default: return set_err(-3, "da_fill: bad dtype %d", dtype);
+        }
+        DA_CHECK_HIP(hipGetLastError());
+        return 0;
+    }
     switch (dtype) {
     case DA_F64:
-        hipLaunchKernelGGL(fill_kernel<double>, dim3(g), dim3(TPB), 0, s,
-                           (double*)chunk, v, n); break;
+        hipLaunchKernelGGL((fill_kernel<double, false>), dim3(g), dim3(TPB),
+                           0, s, (double*)chunk, v, n); break;
     case DA_F32:
-        hipLaunchKernelGGL(fill_kernel<float>, dim3(g), dim3(TPB), 0, s,
-                           (float*)chunk, (float)v, n); break;
+        hipLaunchKernelGGL((fill_kernel<float, false>), dim3(g), dim3(TPB),
+                           0, s, (float*)chunk, (float)v, n); break;
     case DA_I64:
-        hipLaunchKernelGGL(fill_kernel<int64_t>, dim3(g), dim3(TPB), 0, s,
-                           (int64_t*)chunk, (int64_t)v, n); break;
+        hipLaunchKernelGGL((fill_kernel<int64_t, false>), dim3(g), dim3(TPB),
+                           0, s, (int64_t*)chunk, (int64_t)v, n); break;
     default: return set_err(-3, "da_fill: bad dtype %d", dtype);
     }
     DA_CHECK_HIP(hipGetLastError());
@@ -394,7 +427,7 @@ int launch_map2(int opcode, void* dst, const void* a, const void* b,
 }
 
 // ------------------------------------------- fused broadcast & BLAS-1 like
-template <typename T>
+template <typename T, bool NT>
 __global__ void bcast_fma_kernel(T* __restrict__ d, const T* __restrict__ a,
                                  const T* __restrict__ b, T c, uint64_t n) {
     uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
@@ -408,7 +441,8 @@ __global__ void bcast_fma_kernel(T* __restrict__ d, const T* __restrict__ a,
         V x = av[j], y = bv[j], r;
         r.x = x.x * y.x + c;   // -ffp-contract=off: mul then add (Julia Base)
         r.y = x.y * y.y + c;
-        dv[j] = r;
+        if (NT) __builtin_nontemporal_store(r, dv + j);
+        else dv[j] = r;
     }
     for (uint64_t j = 2 * nv + i; j < n; j += stride) d[j] = a[j] * b[j] + c;
 }
@@ -417,12 +451,26 @@ int launch_bcast_fma(void* d, const void* a, const void* b, double c,
                      uint64_t n, int dtype, hipStream_t s) {
     if (n == 0) return 0;
     int g = nblocks(n / 2 + 1);
+    if (use_nt()) {
+        switch (dtype) {
+        case DA_F64: hipLaunchKernelGGL((bcast_fma_kernel<double, true>),
+                        dim3(g), dim3(TPB), 0, s, (double*)d,
+                        (const double*)a, (const double*)b, c, n); break;
+        case DA_F32: hipLaunchKernelGGL((bcast_fma_kernel<float, true>),
+                        dim3(g), dim3(TPB), 0, s, (float*)d,
+                        (const float*)a, (const float*)b, (float)c, n);
+            break;
+        default: return set_err(-3, "da_bcast_fma: bad dtype %d", dtype);
+        }
+        DA_CHECK_HIP(hipGetLastError());
+        return 0;
+    }
     switch (dtype) {
-    case DA_F64: hipLaunchKernelGGL(bcast_fma_kernel<double>, dim3(g),
-                    dim3(TPB), 0, s, (double*)d, (const double*)a,
+    case DA_F64: hipLaunchKernelGGL((bcast_fma_kernel<double, false>),
+                    dim3(g), dim3(TPB), 0, s, (double*)d, (const double*)a,
                     (const double*)b, c, n); break;
-    case DA_F32: hipLaunchKernelGGL(bcast_fma_kernel<float>, dim3(g),
-                    dim3(TPB), 0, s, (float*)d, (const float*)a,
+    case DA_F32: hipLaunchKernelGGL((bcast_fma_kernel<float, false>),
+                    dim3(g), dim3(TPB), 0, s, (float*)d, (const float*)a,
                     (const float*)b, (float)c, n); break;
     default: return set_err(-3, "da_bcast_fma: bad dtype %d", dtype);
     }
