@@ -20,6 +20,9 @@ def init(device=None, rank=None, nranks=None, uid_path=None):
     """Idempotent; collective when nranks > 1."""
     if _state["inited"]:
         return rank_info()
+    # dmabuf IPC is the only mode the host driver supports (env note in
+    # the build environment); harmless if already exported
+    os.environ.setdefault("HSA_ENABLE_IPC_MODE_LEGACY", "0")
     if rank is None:
         rank = int(os.environ.get("RANK", "0"))
     if nranks is None:

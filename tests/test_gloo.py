@@ -18,10 +18,10 @@ from oracle import philox, ops as oops
 WORLD = 2
 
 
-def _init(rank, tmpfile):
+def _init(rank, tmpfile, world=WORLD):
     os.environ["MASTER_ADDR"] = "127.0.0.1"
     td.init_process_group("gloo", init_method="file://%s" % tmpfile,
-                          rank=rank, world_size=WORLD)
+                          rank=rank, world_size=world)
 
 
 def _reduce_worker(rank, tmpfile, q):
@@ -54,16 +54,17 @@ def _reduce_worker(rank, tmpfile, q):
             td.destroy_process_group()
 
 
-def _matmul_worker(rank, tmpfile, q):
+def _matmul_worker(rank, tmpfile, q, world=WORLD):
     try:
-        _init(rank, tmpfile)
+        _init(rank, tmpfile, world)
+        WORLD_ = world
         m, kk, n = 64, 48, 32
         A = np.asfortranarray(philox.fill_uniform_f64(m * kk, 1)
                               .reshape(m, kk, order="F"))
         B = np.asfortranarray(philox.fill_uniform_f64(kk * n, 2)
                               .reshape(kk, n, order="F"))
-        A_dist = tuple(pg.defaultdist((m, kk), WORLD))
-        B_dist = tuple(pg.defaultdist((kk, n), WORLD))
+        A_dist = tuple(pg.defaultdist((m, kk), WORLD_))
+        B_dist = tuple(pg.defaultdist((kk, n), WORLD_))
         A_idxs, A_cuts = pg.chunk_indices((m, kk), A_dist)
         B_idxs, B_cuts = pg.chunk_indices((kk, n), B_dist)
         I, J = A_dist
@@ -137,7 +138,7 @@ def _matmul_worker(rank, tmpfile, q):
             src = i + I * jj
             C_loc += partials[myk] if src == rank else got[(src, myk)]
 
-        gathered = [None] * WORLD
+        gathered = [None] * WORLD_
         td.all_gather_object(gathered, C_loc)
         C = np.zeros((m, n), order="F")
         for r in range(I * K):
@@ -155,15 +156,15 @@ def _matmul_worker(rank, tmpfile, q):
             td.destroy_process_group()
 
 
-def _spawn(fn, tmp_path):
+def _spawn(fn, tmp_path, world=WORLD, extra=()):
     ctx = mp.get_context("spawn")
     q = ctx.Queue()
     tmpfile = str(tmp_path / "rdv")
-    procs = [ctx.Process(target=fn, args=(r, tmpfile, q))
-             for r in range(WORLD)]
+    procs = [ctx.Process(target=fn, args=(r, tmpfile, q) + tuple(extra))
+             for r in range(world)]
     for p in procs:
         p.start()
-    results = [q.get(timeout=180) for _ in range(WORLD)]
+    results = [q.get(timeout=180) for _ in range(world)]
     for p in procs:
         p.join(timeout=60)
     for rank, ok, err in results:
@@ -178,3 +179,9 @@ def test_gloo_reduce_world2(tmp_path):
 @pytest.mark.timeout(300)
 def test_gloo_matmul_world2(tmp_path):
     _spawn(_matmul_worker, tmp_path)
+
+
+@pytest.mark.timeout(300)
+def test_gloo_matmul_world4(tmp_path):
+    # 2x2 grid: real B-panel all-to-all AND partial exchange (K=2)
+    _spawn(_matmul_worker, tmp_path, world=4, extra=(4,))
