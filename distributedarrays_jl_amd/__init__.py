@@ -24,7 +24,8 @@ from .ops import (map_, dmap, map2_, elementwise, broadcast_fma, axpy_,
                   add_, scale_, mapreduce, dsum, dprod, dmaximum, dminimum,
                   dextrema, dmean, ddot, dnorm, dmatmul, dreduce_dims,
                   dsum_dims, dprod_dims, dmaximum_dims, dminimum_dims,
-                  dmean_dims, dmatvec)
+                  dmean_dims, dmatvec, gather_box, map_general,
+                  broadcast_fma_general)
 
 __all__ = [
     "DArray", "DArrayError", "comm", "geometry", "plan",
@@ -35,4 +36,5 @@ __all__ = [
     "dminimum", "dextrema", "dmean", "ddot", "dnorm", "dmatmul",
     "dreduce_dims", "dsum_dims", "dprod_dims", "dmaximum_dims",
     "dminimum_dims", "dmean_dims", "dmatvec",
+    "gather_box", "map_general", "broadcast_fma_general",
 ]

@@ -549,3 +549,138 @@ def dmatvec(A, x, alpha=1.0):
     for buf in recvs.values():
         buf.free()
     return y
+
+
+# ------------------------------------------------- makelocal halo gather
+def gather_box(A, boxes_all):
+    """Collective makelocal (darray.jl:351-368): every rank passes the
+    SAME boxes_all list (boxes_all[r] = the box rank r requests, derived
+    from shared metadata, or None) and receives its own box gathered
+    into a device buffer (column-major).  Returns (_Buf, shape) or
+    (None, None) when this rank requested nothing.  1-D/2-D only.
+
+    Fully-local requests degrade to one on-device strided copy (the
+    reference's zero-copy view case); remote pieces move as grouped
+    ncclSend/Recv of packed sub-blocks over xGMI."""
+    nd = A.ndims
+    if nd > 2:
+        raise DArrayError("gather_box: 1-D/2-D only (round 1)")
+    esz = DTYPE_SIZE[A.dtype]
+    mybox = boxes_all[A.rank] if A.rank < len(boxes_all) else None
+    pieces = plan.halo_plan(A.idxs, A.ranks, boxes_all)
+    me = A.rank
+    my_sends = [p for p in pieces if p[0] == me and p[1] != me]
+    my_recvs = [p for p in pieces if p[1] == me and p[0] != me]
+    my_local = [p for p in pieces if p[0] == me and p[1] == me]
+
+    out = None
+    oshape = None
+    if mybox is not None:
+        oshape = tuple(hi - lo for lo, hi in mybox)
+        out = _Buf(max(geometry.nelems(mybox), 1) * esz)
+
+    def box2d(box):
+        if nd == 1:
+            return box[0], (0, 1)
+        return box[0], box[1]
+
+    def nelems(box):
+        n = 1
+        for lo, hi in box:
+            n *= hi - lo
+        return n
+
+    sendbufs = []
+    for (src, dst, box) in my_sends:
+        (rlo, rhi), (clo, chi) = box2d(box)
+        (llo, _), (lco, _) = box2d(A.lidx)
+        buf = _Buf(nelems(box) * esz)
+        srows = A.lshape[0]
+        off = ((rlo - llo) + (clo - lco) * srows) * esz
+        _copy2d(buf.p, (rhi - rlo) * esz,
+                ctypes.c_void_p(A._ptr().value + off), srows * esz,
+                (rhi - rlo) * esz, chi - clo)
+        sendbufs.append((box, buf, dst))
+    recvbufs = []
+    for (src, dst, box) in my_recvs:
+        recvbufs.append((box, _Buf(nelems(box) * esz), src))
+    if my_sends or my_recvs:
+        check(lib.da_group_start())
+        for box, buf, dst in sendbufs:
+            check(lib.da_send(buf.p, nelems(box) * esz, dst))
+        for box, buf, src in recvbufs:
+            check(lib.da_recv(buf.p, nelems(box) * esz, src))
+        check(lib.da_group_end())
+    if mybox is not None:
+        (mlo, _), (mco, _) = box2d(mybox)
+        orows = oshape[0]
+        for (src, dst, box) in my_local:
+            (rlo, rhi), (clo, chi) = box2d(box)
+            (llo, _), (lco, _) = box2d(A.lidx)
+            srows = A.lshape[0]
+            soff = ((rlo - llo) + (clo - lco) * srows) * esz
+            doff = ((rlo - mlo) + (clo - mco) * orows) * esz
+            _copy2d(out.at(doff), orows * esz,
+                    ctypes.c_void_p(A._ptr().value + soff), srows * esz,
+                    (rhi - rlo) * esz, chi - clo)
+        for box, buf, src in recvbufs:
+            (rlo, rhi), (clo, chi) = box2d(box)
+            doff = ((rlo - mlo) + (clo - mco) * orows) * esz
+            _copy2d(out.at(doff), orows * esz, buf.p, (rhi - rlo) * esz,
+                    (rhi - rlo) * esz, chi - clo)
+    check(lib.da_synchronize())
+    for _, buf, _ in sendbufs:
+        buf.free()
+    for _, buf, _ in recvbufs:
+        buf.free()
+    return out, oshape
+
+
+def _dest_boxes(dest):
+    """boxes_all for 'each dest owner requests its own index box' — the
+    makelocal(src, localindices(dest)) pattern (mapreduce.jl:8,
+    broadcast.jl:79, linalg.jl:42)."""
+    boxes = [None] * dest.nranks
+    for c, r in enumerate(dest.ranks):
+        boxes[r] = dest.idxs[c]
+    return boxes
+
+
+def map_general(op, dest, src):
+    """map!(f, dest, src) for MISMATCHED cuts (mapreduce.jl:5-12 with
+    makelocal): gather src's piece of dest's index box, then map."""
+    if dest.dims != src.dims or dest.dtype != src.dtype:
+        raise DArrayError("map_general: dims/dtype mismatch")
+    if dest.dist == src.dist and dest.ranks == src.ranks:
+        return map_(op, dest, src)
+    buf, shape = gather_box(src, _dest_boxes(dest))
+    if buf is not None:
+        check(lib.da_map(MAP_OP[op], dest._ptr(), buf.p, dest.lnumel,
+                         DTYPES[dest.dtype]))
+        check(lib.da_synchronize())
+        buf.free()
+    return dest
+
+
+def broadcast_fma_general(dest, a, b, c):
+    """D .= A .* B .+ c with arbitrary (mismatched) cuts — the
+    bclocal/makelocal localisation of broadcast.jl:65-85."""
+    if dest.samedist(a) and dest.samedist(b):
+        return broadcast_fma(dest, a, b, c)
+    boxes = _dest_boxes(dest)
+    abuf = bbuf = None
+    if not dest.samedist(a):
+        abuf, _ = gather_box(a, boxes)
+    if not dest.samedist(b):
+        bbuf, _ = gather_box(b, boxes)
+    ap = abuf.p if abuf is not None else a._ptr()
+    bp = bbuf.p if bbuf is not None else b._ptr()
+    if dest.lnumel:
+        check(lib.da_bcast_fma(dest._ptr(), ap, bp, float(c),
+                               dest.lnumel, DTYPES[dest.dtype]))
+    check(lib.da_synchronize())
+    if abuf is not None:
+        abuf.free()
+    if bbuf is not None:
+        bbuf.free()
+    return dest

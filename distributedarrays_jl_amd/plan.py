@@ -81,6 +81,33 @@ def partial_plan(A_dist, K):
     return moves
 
 
+def halo_plan(src_idxs, src_ranks, boxes):
+    """Pieces (src_rank, dst_rank, box) for the makelocal halo gather
+    (darray.jl:351-368: requested indices not fully local are fetched
+    from the owning chunks; here via grouped ncclSend/Recv).
+
+    src_idxs: chunk boxes of the source DArray; src_ranks[c] = owner.
+    boxes[r]: the box rank r requests (None = no request).  Returns a
+    deterministic sorted list in GLOBAL coordinates."""
+    pieces = []
+    for dst, box in enumerate(boxes):
+        if box is None:
+            continue
+        for c, idx in enumerate(src_idxs):
+            inter = []
+            ok = True
+            for (blo, bhi), (clo, chi) in zip(box, idx):
+                lo, hi = max(blo, clo), min(bhi, chi)
+                if hi <= lo:
+                    ok = False
+                    break
+                inter.append((lo, hi))
+            if ok:
+                pieces.append((src_ranks[c], dst, tuple(inter)))
+    pieces.sort()
+    return pieces
+
+
 def accumulate_order(J):
     """Owner-side add! order over j (ascending — one valid schedule of
     the reference's async accumulation, linalg.jl:243-251)."""
