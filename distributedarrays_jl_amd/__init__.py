@@ -16,7 +16,7 @@ imported here.
 """
 
 from ._ffi import DArrayError
-from . import comm, geometry, plan
+from . import comm, geometry, plan, spmd
 from .darray import (DArray, dzeros, dones, dfill, drand, drandn,
                      distribute, localpart, localindices, d_closeall,
                      bytes_in_use)
@@ -28,7 +28,7 @@ from .ops import (map_, dmap, map2_, elementwise, broadcast_fma, axpy_,
                   broadcast_fma_general)
 
 __all__ = [
-    "DArray", "DArrayError", "comm", "geometry", "plan",
+    "DArray", "DArrayError", "comm", "geometry", "plan", "spmd",
     "dzeros", "dones", "dfill", "drand", "drandn", "distribute",
     "localpart", "localindices", "d_closeall", "bytes_in_use",
     "map_", "dmap", "map2_", "elementwise", "broadcast_fma", "axpy_",
