@@ -25,7 +25,7 @@ from .ops import (map_, dmap, map2_, elementwise, broadcast_fma, axpy_,
                   dextrema, dmean, ddot, dnorm, dmatmul, dreduce_dims,
                   dsum_dims, dprod_dims, dmaximum_dims, dminimum_dims,
                   dmean_dims, dmatvec, gather_box, map_general,
-                  broadcast_fma_general)
+                  broadcast_fma_general, dsort)
 
 __all__ = [
     "DArray", "DArrayError", "comm", "geometry", "plan", "spmd",
@@ -36,5 +36,5 @@ __all__ = [
     "dminimum", "dextrema", "dmean", "ddot", "dnorm", "dmatmul",
     "dreduce_dims", "dsum_dims", "dprod_dims", "dmaximum_dims",
     "dminimum_dims", "dmean_dims", "dmatvec",
-    "gather_box", "map_general", "broadcast_fma_general",
+    "gather_box", "map_general", "broadcast_fma_general", "dsort",
 ]

@@ -58,6 +58,9 @@ _sigs = {
     "da_scale": ([ptr, f64, u64, i32], i32),
     "da_reduce": ([i32, i32, ptr, u64, i32, ptr], i32),
     "da_reduce_dims": ([i32, i32, ptr, u64, u64, u64, i32, ptr], i32),
+    "da_sort": ([ptr, u64, i32], i32),
+    "da_lower_bound": ([ptr, u64, i32, ptr, i32,
+                        ctypes.POINTER(ctypes.c_uint64)], i32),
     "da_allreduce": ([ptr, i32, i32, i32], i32),
     "da_gemm_f64": ([ptr, ptr, ptr, i64, i64, i64, i64, i64, i64, f64, f64],
                     i32),

@@ -92,6 +92,34 @@ class DArray:
                                  % (arr.shape, self.lshape))
             self.set_localpart(arr)
 
+    @classmethod
+    def from_chunk_sizes(cls, sizes, dtype="f64"):
+        """1-D DArray with explicit (possibly ragged) per-rank chunk
+        sizes — the samplesort result shape (sort.jl returns a DArray
+        whose distribution follows the splitter boundaries)."""
+        _auto_init()
+        rank, nr = comm.rank_info()
+        if len(sizes) != nr:
+            raise ValueError("need one size per rank")
+        total = sum(sizes)
+        d = cls((total,), dtype, (nr,), _alloc=False)
+        # override the even cuts with the ragged ones (1-based, ref style)
+        cuts = [1]
+        for s in sizes:
+            cuts.append(cuts[-1] + int(s))
+        d.cuts = [cuts]
+        d.idxs = [((cuts[i] - 1, cuts[i + 1] - 1),) for i in range(nr)]
+        d.lchunk = rank
+        d.lidx = d.idxs[rank]
+        d.lshape = (int(sizes[rank]),)
+        d.lnumel = int(sizes[rank])
+        p = ctypes.c_void_p()
+        check(lib.da_alloc(max(d.lnumel, 1) * DTYPE_SIZE[dtype],
+                           DTYPES[dtype], ctypes.byref(p)))
+        d._chunk = p
+        _registry[d.id] = d
+        return d
+
     # ---- lifetime (darray.jl:46-49 -> core.jl:67-103) ----
     def close(self):
         if self._chunk is not None:
@@ -138,6 +166,9 @@ class DArray:
         if self._chunk is None:
             raise _ffi.DArrayError("use after close (darray id %d)" % self.id)
         return self._chunk
+
+    def at_byte(self, off):
+        return ctypes.c_void_p(self._ptr().value + int(off))
 
     # ---- localpart access (darray.jl:330-337, :394-400) ----
     def localindices(self):

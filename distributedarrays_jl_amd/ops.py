@@ -684,3 +684,111 @@ def broadcast_fma_general(dest, a, b, c):
     if bbuf is not None:
         bbuf.free()
     return dest
+
+
+# ------------------------------------------------------------ samplesort
+def dsort(d, samples_per_rank=64):
+    """sort(::DVector) — the distributed samplesort of src/sort.jl:103-170,
+    MI355X-native: per-chunk rocPRIM radix sort, evenly-spaced samples
+    gathered to agree on P-1 splitters (control plane; the reference
+    gathers samples to the caller the same way), device binary-search
+    boundaries, all-to-all of the contiguous sorted segments over xGMI,
+    and a local re-sort of the received runs.  Returns a new DVector
+    whose (possibly ragged) distribution follows the splitters."""
+    import numpy as np
+    if d.ndims != 1:
+        raise DArrayError("dsort: DVector only")
+    if d.dtype == "f32":
+        raise DArrayError("dsort: f64/i64 only (round 1)")
+    esz = DTYPE_SIZE[d.dtype]
+    npdt = np.dtype(NUMPY_DTYPES[d.dtype])
+    P = d.nranks
+    out = d.copy()
+    if out.lnumel > 1:
+        check(lib.da_sort(out._ptr(), out.lnumel, DTYPES[d.dtype]))
+    if P == 1:
+        return out
+
+    import torch.distributed as td
+    if not td.is_initialized():
+        raise DArrayError("dsort with nranks>1 needs gloo control plane")
+
+    # evenly spaced samples from the sorted chunk (sort.jl sampling)
+    n_loc = out.lnumel
+    s = min(samples_per_rank, max(n_loc, 0))
+    samples = np.empty(s, dtype=npdt)
+    if s:
+        idx = ((np.arange(s) + 0.5) * n_loc / s).astype(np.int64)
+        # strided device->host via copy2d into a packed staging buffer
+        stage = _Buf(max(s, 1) * esz)
+        for t, i in enumerate(idx):   # s is small (<=64): cheap copies
+            check(lib.da_d2d(stage.at(t * esz),
+                             ctypes.c_void_p(out._ptr().value
+                                             + int(i) * esz), esz))
+        check(lib.da_d2h(stage.p, samples.ctypes.data_as(ctypes.c_void_p),
+                         s * esz))
+        stage.free()
+    gathered = [None] * P
+    td.all_gather_object(gathered, samples)
+    allsamp = np.sort(np.concatenate([g for g in gathered if g is not None]))
+    if allsamp.size < P:
+        splitters = allsamp[:max(P - 1, 0)]
+    else:
+        splitters = allsamp[[(i + 1) * allsamp.size // P
+                             for i in range(P - 1)]]
+
+    # boundaries in my sorted chunk
+    bounds = (ctypes.c_uint64 * max(P - 1, 1))()
+    if n_loc and P > 1:
+        spl = np.ascontiguousarray(splitters, dtype=npdt)
+        check(lib.da_lower_bound(out._ptr(), n_loc, DTYPES[d.dtype],
+                                 spl.ctypes.data_as(ctypes.c_void_p),
+                                 P - 1, bounds))
+    edges = [0] + [int(bounds[i]) for i in range(P - 1)] + [n_loc]
+
+    # segment sizes: seg[j] goes to rank j; exchange counts via gloo
+    segs = [edges[j + 1] - edges[j] for j in range(P)]
+    allsegs = [None] * P
+    td.all_gather_object(allsegs, segs)
+    recv_sizes = [allsegs[src][d.rank] for src in range(P)]
+    my_total = sum(recv_sizes)
+
+    res = DArray.from_chunk_sizes(
+        [sum(allsegs[src][dst] for src in range(P)) for dst in range(P)],
+        d.dtype)
+    assert res.lnumel == my_total
+    # grouped all-to-all of contiguous sorted segments
+    recvbufs = {}
+    check(lib.da_group_start())
+    for dst in range(P):
+        if dst != d.rank and segs[dst]:
+            check(lib.da_send(
+                ctypes.c_void_p(out._ptr().value + edges[dst] * esz),
+                segs[dst] * esz, dst))
+    for src in range(P):
+        if src != d.rank and recv_sizes[src]:
+            buf = _Buf(recv_sizes[src] * esz)
+            recvbufs[src] = buf
+            check(lib.da_recv(buf.p, recv_sizes[src] * esz, src))
+    check(lib.da_group_end())
+    # concatenate runs (own + received) then radix re-sort
+    off = 0
+    for src in range(P):
+        nsz = recv_sizes[src]
+        if not nsz:
+            continue
+        if src == d.rank:
+            check(lib.da_d2d(res.at_byte(off),
+                             ctypes.c_void_p(out._ptr().value
+                                             + edges[d.rank] * esz),
+                             nsz * esz))
+        else:
+            check(lib.da_d2d(res.at_byte(off), recvbufs[src].p, nsz * esz))
+        off += nsz * esz
+    if res.lnumel > 1:
+        check(lib.da_sort(res._ptr(), res.lnumel, DTYPES[d.dtype]))
+    check(lib.da_synchronize())
+    for buf in recvbufs.values():
+        buf.free()
+    out.close()
+    return res

@@ -132,6 +132,11 @@ int da_gemm_f64(void* C, const void* A, const void* B,
                 int64_t lda, int64_t ldb, int64_t ldc,
                 double alpha, double beta);
 
+/* ---- distributed samplesort building blocks (src/sort.jl:103-170) --- */
+int da_sort(void* chunk, uint64_t n, int dtype);       /* per-chunk radix sort */
+int da_lower_bound(const void* sorted, uint64_t n, int dtype,
+                   const void* splitters, int k, uint64_t* out);
+
 /* ---- point-to-point (panel / halo exchange; replaces the remotecall
  *      shipping of B panels and partial products, linalg.jl:211-251,
  *      and makelocal's remote fetch, darray.jl:351-368) ----------------- */
