@@ -736,6 +736,10 @@ def dsort(d, samples_per_rank=64):
     else:
         splitters = allsamp[[(i + 1) * allsamp.size // P
                              for i in range(P - 1)]]
+    if splitters.size < P - 1:   # tiny inputs: trailing segments empty
+        pad = (np.inf if npdt.kind == "f" else np.iinfo(npdt).max)
+        splitters = np.concatenate(
+            [splitters, np.full(P - 1 - splitters.size, pad, npdt)])
 
     # boundaries in my sorted chunk
     bounds = (ctypes.c_uint64 * max(P - 1, 1))()

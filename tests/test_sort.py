@@ -26,6 +26,11 @@ def simulate_samplesort(x, P, s=64):
     else:
         splitters = allsamp[[(i + 1) * allsamp.size // P
                              for i in range(P - 1)]]
+    if splitters.size < P - 1:   # tiny inputs: trailing buckets empty
+        pad = (np.inf if x.dtype.kind == "f"
+               else np.iinfo(x.dtype).max)
+        splitters = np.concatenate(
+            [splitters, np.full(P - 1 - splitters.size, pad, x.dtype)])
     buckets = [[] for _ in range(P)]
     for ch in chunks:
         edges = [0] + [int(np.searchsorted(ch, sp, side="left"))
