@@ -185,3 +185,9 @@ def test_gloo_matmul_world2(tmp_path):
 def test_gloo_matmul_world4(tmp_path):
     # 2x2 grid: real B-panel all-to-all AND partial exchange (K=2)
     _spawn(_matmul_worker, tmp_path, world=4, extra=(4,))
+
+
+@pytest.mark.timeout(300)
+def test_gloo_matmul_world8(tmp_path):
+    # 2x4 grid — the exact cfg-4 bench geometry (16384^2 over 8 GPUs)
+    _spawn(_matmul_worker, tmp_path, world=8, extra=(8,))
