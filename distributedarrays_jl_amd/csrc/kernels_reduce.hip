@@ -201,6 +201,28 @@ __global__ void reduce_dims_waves(int mapop, int redop,
     }
 }
 
+// Variant C: one 256-thread block per output element — for few outputs
+// over a long axis (e.g. sum(D, dims=2) of a square matrix), where the
+// thread/wave variants leave the chip underfilled (measured 137 GB/s vs
+// 3+ TB/s; profiles/r01_kernel_stats.md).
+template <typename T>
+__global__ void reduce_dims_blocks(int mapop, int redop,
+                                   const T* __restrict__ src,
+                                   uint64_t inner, uint64_t axis,
+                                   uint64_t outer, T* __restrict__ dst) {
+    uint64_t total = inner * outer;
+    for (uint64_t e = blockIdx.x; e < total; e += gridDim.x) {
+        uint64_t i = e % inner, o = e / inner;
+        const T* p = src + i + o * inner * axis;
+        T acc = RedIdent<T>::get(redop);
+        for (uint64_t a = threadIdx.x; a < axis; a += blockDim.x)
+            acc = red_comb(redop, acc, mapf(mapop, p[a * inner]));
+        acc = block_reduce(redop, acc);
+        if (threadIdx.x == 0) dst[i + o * inner] = acc;
+        __syncthreads();   // LDS in block_reduce reused next iteration
+    }
+}
+
 template <typename T>
 static int do_reduce_dims(int mapop, int redop, const T* src,
                           uint64_t inner, uint64_t axis, uint64_t outer,
@@ -209,15 +231,23 @@ static int do_reduce_dims(int mapop, int redop, const T* src,
     if (total == 0) return 0;
     // axis == 0 still runs: the loop body never executes and dst gets
     // the fold identity.
-    if (inner >= 64) {
+    if (total >= 262144 && inner >= 64) {
+        // plenty of outputs, coalesced across inner: thread-per-output
         uint64_t want = (total + RTPB - 1) / RTPB;
         int g = (int)(want < 1 ? 1 : (want > 8192 ? 8192 : want));
         hipLaunchKernelGGL(reduce_dims_threads<T>, dim3(g), dim3(RTPB), 0,
                            s, mapop, redop, src, inner, axis, outer, dst);
-    } else {
+    } else if (inner < 64 && axis >= 64 && total >= 16384) {
+        // inner tiny (reduce over leading dim): wave-per-output, lanes
+        // stride the axis (coalesced along the axis)
         uint64_t want = (total * 64 + RTPB - 1) / RTPB;
         int g = (int)(want < 1 ? 1 : (want > 8192 ? 8192 : want));
         hipLaunchKernelGGL(reduce_dims_waves<T>, dim3(g), dim3(RTPB), 0,
+                           s, mapop, redop, src, inner, axis, outer, dst);
+    } else {
+        // few outputs / long axis: block-per-output
+        int g = (int)(total < 1 ? 1 : (total > 8192 ? 8192 : total));
+        hipLaunchKernelGGL(reduce_dims_blocks<T>, dim3(g), dim3(RTPB), 0,
                            s, mapop, redop, src, inner, axis, outer, dst);
     }
     DA_CHECK_HIP(hipGetLastError());
