@@ -223,6 +223,44 @@ __global__ void reduce_dims_blocks(int mapop, int redop,
     }
 }
 
+// Variant D: few outputs, LARGE inner (e.g. sum(8192x8192, dims=2)):
+// split the axis into NB column slices; each block walks a row tile of
+// its slice with fully coalesced reads (consecutive threads =
+// consecutive rows), producing NB partial rows; a second kernel folds
+// the NB partials per row.
+template <typename T>
+__global__ void reduce_dims_slices1(int mapop, int redop,
+                                    const T* __restrict__ src,
+                                    uint64_t inner, uint64_t axis,
+                                    int nb, T* __restrict__ partials) {
+    uint64_t o = blockIdx.z;
+    const T* base = src + o * inner * axis;
+    uint64_t row = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (row >= inner) return;
+    uint64_t a0 = axis * blockIdx.y / nb;
+    uint64_t a1 = axis * (blockIdx.y + 1) / nb;
+    T acc = RedIdent<T>::get(redop);
+    for (uint64_t a = a0; a < a1; ++a)
+        acc = red_comb(redop, acc, mapf(mapop, base[row + a * inner]));
+    partials[(o * nb + blockIdx.y) * inner + row] = acc;
+}
+
+template <typename T>
+__global__ void reduce_dims_slices2(int redop, const T* __restrict__ partials,
+                                    uint64_t inner, int nb,
+                                    T* __restrict__ dst, uint64_t outer) {
+    uint64_t e = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t t = e; t < inner * outer; t += stride) {
+        uint64_t row = t % inner, o = t / inner;
+        T acc = RedIdent<T>::get(redop);
+        for (int b = 0; b < nb; ++b)
+            acc = red_comb(redop, acc,
+                           partials[(o * nb + b) * inner + row]);
+        dst[row + o * inner] = acc;
+    }
+}
+
 template <typename T>
 static int do_reduce_dims(int mapop, int redop, const T* src,
                           uint64_t inner, uint64_t axis, uint64_t outer,
@@ -231,6 +269,27 @@ static int do_reduce_dims(int mapop, int redop, const T* src,
     if (total == 0) return 0;
     // axis == 0 still runs: the loop body never executes and dst gets
     // the fold identity.
+    if (total < 262144 && inner >= 64 && axis >= 64 && outer <= 64) {
+        // fill ~2048 workgroups: (inner/256 row tiles) x nb slices
+        uint64_t row_tiles = (inner + RTPB - 1) / RTPB;
+        int nb = (int)(2048 / row_tiles);
+        if (nb < 1) nb = 1;
+        if (nb > 64) nb = 64;
+        if ((uint64_t)nb > axis) nb = (int)axis;
+        int rc = ensure_partials((uint64_t)nb * inner * outer * sizeof(T));
+        if (rc) return rc;
+        T* parts = (T*)st().partials;
+        dim3 g((inner + RTPB - 1) / RTPB, nb, outer);
+        hipLaunchKernelGGL(reduce_dims_slices1<T>, g, dim3(RTPB), 0, s,
+                           mapop, redop, src, inner, axis, nb, parts);
+        DA_CHECK_HIP(hipGetLastError());
+        uint64_t want = (total + RTPB - 1) / RTPB;
+        int g2 = (int)(want < 1 ? 1 : (want > 2048 ? 2048 : want));
+        hipLaunchKernelGGL(reduce_dims_slices2<T>, dim3(g2), dim3(RTPB), 0,
+                           s, redop, parts, inner, nb, dst, outer);
+        DA_CHECK_HIP(hipGetLastError());
+        return 0;
+    }
     if (total >= 262144 && inner >= 64) {
         // plenty of outputs, coalesced across inner: thread-per-output
         uint64_t want = (total + RTPB - 1) / RTPB;
