@@ -273,30 +273,27 @@ void gemm_f64_mfma_v3(const double* __restrict__ A,
         for (int j = 0; j < 2; ++j) acc[i][j] = {0.0, 0.0, 0.0, 0.0};
 
     // staging: A/B tiles are 2048 doubles each; 512 threads x 2 f64x2
-    int a_c[2], a_r2[2], b_n[2], b_k2[2];
-#pragma unroll
-    for (int r = 0; r < 2; ++r) {
-        int idx2 = tid + r * 512;
-        a_c[r] = idx2 >> 6;
-        a_r2[r] = (idx2 & 63) * 2;
-        b_n[r] = idx2 >> 3;
-        b_k2[r] = (idx2 & 7) * 2;
-    }
+    // (coordinates recomputed inline — keeping them in arrays costs
+    // VGPRs against the __launch_bounds__(512,4) budget)
+#define A_C(r)  ((tid + (r) * 512) >> 6)
+#define A_R2(r) (((tid + (r) * 512) & 63) * 2)
+#define B_N(r)  ((tid + (r) * 512) >> 3)
+#define B_K2(r) (((tid + (r) * 512) & 7) * 2)
 
     f64x2 pa[2], pb[2];
     const int64_t ktiles = k / BK;
 #pragma unroll
     for (int r = 0; r < 2; ++r) {
-        pa[r] = *reinterpret_cast<const f64x2*>(A + (int64_t)a_c[r] * lda
-                                                + bm + a_r2[r]);
-        pb[r] = *reinterpret_cast<const f64x2*>(B + (bn + b_n[r]) * ldb
-                                                + b_k2[r]);
+        pa[r] = *reinterpret_cast<const f64x2*>(A + (int64_t)A_C(r) * lda
+                                                + bm + A_R2(r));
+        pb[r] = *reinterpret_cast<const f64x2*>(B + (bn + B_N(r)) * ldb
+                                                + B_K2(r));
     }
 #pragma unroll
     for (int r = 0; r < 2; ++r) {
-        As[(a_r2[r] + 0) * LSTR + a_c[r]] = pa[r].x;
-        As[(a_r2[r] + 1) * LSTR + a_c[r]] = pa[r].y;
-        *reinterpret_cast<f64x2*>(&Bs[b_n[r] * LSTR + b_k2[r]]) = pb[r];
+        As[(A_R2(r) + 0) * LSTR + A_C(r)] = pa[r].x;
+        As[(A_R2(r) + 1) * LSTR + A_C(r)] = pa[r].y;
+        *reinterpret_cast<f64x2*>(&Bs[B_N(r) * LSTR + B_K2(r)]) = pb[r];
     }
 
     for (int64_t kt = 0; kt < ktiles; ++kt) {
@@ -306,9 +303,9 @@ void gemm_f64_mfma_v3(const double* __restrict__ A,
 #pragma unroll
             for (int r = 0; r < 2; ++r) {
                 pa[r] = *reinterpret_cast<const f64x2*>(
-                    A + (k0 + a_c[r]) * lda + bm + a_r2[r]);
+                    A + (k0 + A_C(r)) * lda + bm + A_R2(r));
                 pb[r] = *reinterpret_cast<const f64x2*>(
-                    B + (bn + b_n[r]) * ldb + k0 + b_k2[r]);
+                    B + (bn + B_N(r)) * ldb + k0 + B_K2(r));
             }
         }
 #pragma unroll
@@ -332,13 +329,17 @@ void gemm_f64_mfma_v3(const double* __restrict__ A,
         if (kt + 1 < ktiles) {
 #pragma unroll
             for (int r = 0; r < 2; ++r) {
-                As[(a_r2[r] + 0) * LSTR + a_c[r]] = pa[r].x;
-                As[(a_r2[r] + 1) * LSTR + a_c[r]] = pa[r].y;
-                *reinterpret_cast<f64x2*>(&Bs[b_n[r] * LSTR + b_k2[r]])
+                As[(A_R2(r) + 0) * LSTR + A_C(r)] = pa[r].x;
+                As[(A_R2(r) + 1) * LSTR + A_C(r)] = pa[r].y;
+                *reinterpret_cast<f64x2*>(&Bs[B_N(r) * LSTR + B_K2(r)])
                     = pb[r];
             }
         }
     }
+#undef A_C
+#undef A_R2
+#undef B_N
+#undef B_K2
 
 #pragma unroll
     for (int i = 0; i < 4; ++i) {
