@@ -276,6 +276,13 @@ int da_map2(int opcode, void* dst, const void* a, const void* b,
     return launch_map2(opcode, dst, a, b, n, dtype, st().stream);
 }
 
+int da_map2_scalar(int opcode, void* dst, const void* src, double c,
+                   int reverse, uint64_t n, int dtype) {
+    DA_REQUIRE_INIT();
+    return launch_map2_scalar(opcode, dst, src, c, reverse, n, dtype,
+                              st().stream);
+}
+
 int da_bcast_fma(void* d, const void* a, const void* b, double c,
                  uint64_t n, int dtype) {
     DA_REQUIRE_INIT();

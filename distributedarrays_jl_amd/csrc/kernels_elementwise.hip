@@ -426,6 +426,52 @@ int launch_map2(int opcode, void* dst, const void* a, const void* b,
     return 0;
 }
 
+// scalar-broadcast forms: D .= f.(src, c) / f.(c, src) (e.g. D .+ 1,
+// the cfg-1 plumbing op; Base broadcast with a scalar argument —
+// broadcast.jl:124-133 treats singletons as local, no distribution)
+template <typename T>
+__global__ void map2s_kernel(int op, T* __restrict__ dst,
+                             const T* __restrict__ a, T c, int rev,
+                             uint64_t n) {
+    uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t j = i; j < n; j += stride)
+        dst[j] = rev ? apply_map2<T>(op, c, a[j])
+                     : apply_map2<T>(op, a[j], c);
+}
+
+__global__ void map2s_kernel_i64(int op, int64_t* __restrict__ dst,
+                                 const int64_t* __restrict__ a, int64_t c,
+                                 int rev, uint64_t n) {
+    uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t j = i; j < n; j += stride)
+        dst[j] = rev ? apply_map2_i64(op, c, a[j])
+                     : apply_map2_i64(op, a[j], c);
+}
+
+int launch_map2_scalar(int opcode, void* dst, const void* src, double c,
+                       int rev, uint64_t n, int dtype, hipStream_t s) {
+    if (n == 0) return 0;
+    if (opcode < 0 || opcode >= DA_OP2__N)
+        return set_err(-3, "da_map2_scalar: bad opcode %d", opcode);
+    int g = nblocks(n);
+    switch (dtype) {
+    case DA_F64: hipLaunchKernelGGL(map2s_kernel<double>, dim3(g),
+                    dim3(TPB), 0, s, opcode, (double*)dst,
+                    (const double*)src, c, rev, n); break;
+    case DA_F32: hipLaunchKernelGGL(map2s_kernel<float>, dim3(g),
+                    dim3(TPB), 0, s, opcode, (float*)dst,
+                    (const float*)src, (float)c, rev, n); break;
+    case DA_I64: hipLaunchKernelGGL(map2s_kernel_i64, dim3(g), dim3(TPB),
+                    0, s, opcode, (int64_t*)dst, (const int64_t*)src,
+                    (int64_t)c, rev, n); break;
+    default: return set_err(-3, "da_map2_scalar: bad dtype %d", dtype);
+    }
+    DA_CHECK_HIP(hipGetLastError());
+    return 0;
+}
+
 // ------------------------------------------- fused broadcast & BLAS-1 like
 template <typename T, bool NT>
 __global__ void bcast_fma_kernel(T* __restrict__ d, const T* __restrict__ a,

@@ -58,6 +58,24 @@ def elementwise(op, a, b):
     return map2_(op, a.similar(), a, b)
 
 
+def map2_scalar_(op, dest, src, c, reverse=False):
+    """dest .= f.(src, c) (or f.(c, src)) — scalar broadcast argument
+    (broadcast.jl:124-133: singletons are not distributed).  Covers the
+    cfg-1 plumbing op D .+ 1."""
+    _aligned(dest, src)
+    if dest.dtype == "i64" and op not in I64_MAP2_OPS:
+        raise DArrayError("op %r invalid for i64" % op)
+    if dest.lnumel:
+        check(lib.da_map2_scalar(MAP2_OP[op], dest._ptr(), src._ptr(),
+                                 float(c), 1 if reverse else 0,
+                                 dest.lnumel, DTYPES[dest.dtype]))
+    return dest
+
+
+def elementwise_scalar(op, a, c, reverse=False):
+    return map2_scalar_(op, a.similar(), a, c, reverse)
+
+
 def broadcast_fma(dest, a, b, c):
     """D .= A .* B .+ c — the fused cfg-3 broadcast (broadcast.jl:65-85;
     aligned same-cuts args need zero communication)."""

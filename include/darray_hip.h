@@ -100,6 +100,8 @@ int da_map2(int opcode, void* dst, const void* a, const void* b,
             uint64_t n, int dtype);                          /* elementwise +,-,.. mapreduce.jl:180-189 */
 int da_bcast_fma(void* d, const void* a, const void* b, double c,
                  uint64_t n, int dtype);                     /* D .= A .* B .+ c, broadcast.jl:65-85 */
+int da_map2_scalar(int opcode, void* dst, const void* src, double c,
+                   int reverse, uint64_t n, int dtype);      /* D .+ 1 etc (scalar broadcast arg, broadcast.jl:124-133) */
 int da_axpby(void* y, const void* x, double alpha, double beta,
              uint64_t n, int dtype);                         /* y = alpha*x + beta*y: axpy! linalg.jl:24-34 */
 int da_add(void* dest, const void* src, double scale,

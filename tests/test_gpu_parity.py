@@ -364,3 +364,26 @@ def test_events_api(dja):
     check(lib.da_event_destroy(e0))
     check(lib.da_event_destroy(e1))
     d.close()
+
+
+def test_map2_scalar(dja):
+    x = philox.fill_uniform_f64(40001, seed=61)
+    d = dja.distribute(x)
+    for op, ref in [("add", x + 1.0), ("sub", x - 0.5), ("mul", x * 3.0),
+                    ("div", x / 2.0)]:
+        out = dja.elementwise_scalar(op, d, {"add": 1.0, "sub": 0.5,
+                                             "mul": 3.0, "div": 2.0}[op])
+        assert np.array_equal(out.localpart(), ref), op
+        out.close()
+    # reversed operand order: 2.0 ./ x
+    out = dja.elementwise_scalar("div", d, 2.0, reverse=True)
+    assert np.array_equal(out.localpart(), 2.0 / x)
+    out.close()
+    # i64 scalar
+    with np.errstate(over="ignore"):
+        xi = philox.fill_int64(10001, seed=62)
+        di = dja.distribute(xi)
+        out = dja.elementwise_scalar("add", di, 7)
+        assert np.array_equal(out.localpart(), xi + 7)
+        out.close(); di.close()
+    d.close()
