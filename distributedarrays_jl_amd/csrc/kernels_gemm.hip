@@ -356,6 +356,128 @@ void gemm_f64_mfma_v3(const double* __restrict__ A,
     }
 }
 
+
+// v5: v3 + LDS double-buffering — tile kt lives in buffer kt&1, tile
+// kt+1 is written to the other buffer after the MFMA phase, so each
+// K-tile needs ONE barrier instead of two (the 2-barrier stage+drain
+// pattern is the known ceiling of simple MFMA GEMM loops —
+// cdna_hip_programming.md §5).  LDS 73.7 KB -> still 2 blocks/CU.
+__global__ __launch_bounds__(512, 4)
+void gemm_f64_mfma_v5(const double* __restrict__ A,
+                      const double* __restrict__ B,
+                      double* __restrict__ C, int64_t m, int64_t n,
+                      int64_t k, int64_t lda, int64_t ldb, int64_t ldc,
+                      double alpha, double beta) {
+    __shared__ double lds[2 * 2 * BM * LSTR];   // [buf][A/B][...]
+    // buffer b: A at lds + b*2*BM*LSTR, B at that + BM*LSTR
+
+    const int tid = threadIdx.x;
+    const int wave = tid >> 6;
+    const int lane = tid & 63;
+    const int l16 = lane & 15;
+    const int l4 = lane >> 4;
+    const int wr = (wave >> 2) * 64;
+    const int wc = (wave & 3) * 32;
+
+    const int gx = gridDim.x, nwg = gridDim.x * gridDim.y;
+    int w = blockIdx.y * gx + blockIdx.x;
+    int q = nwg >> 3, rmd = nwg & 7, xcd = w & 7, idx = w >> 3;
+    int sw = (xcd < rmd ? xcd * (q + 1) : rmd * (q + 1) + (xcd - rmd) * q)
+             + idx;
+    const int64_t bm = (int64_t)(sw % gx) * BM;
+    const int64_t bn = (int64_t)(sw / gx) * BN;
+
+    f64x4 acc[4][2];
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+#pragma unroll
+        for (int j = 0; j < 2; ++j) acc[i][j] = {0.0, 0.0, 0.0, 0.0};
+
+#define A_C(r)  ((tid + (r) * 512) >> 6)
+#define A_R2(r) (((tid + (r) * 512) & 63) * 2)
+#define B_N(r)  ((tid + (r) * 512) >> 3)
+#define B_K2(r) (((tid + (r) * 512) & 7) * 2)
+
+    f64x2 pa[2], pb[2];
+    const int64_t ktiles = k / BK;
+#pragma unroll
+    for (int r = 0; r < 2; ++r) {
+        pa[r] = *reinterpret_cast<const f64x2*>(A + (int64_t)A_C(r) * lda
+                                                + bm + A_R2(r));
+        pb[r] = *reinterpret_cast<const f64x2*>(B + (bn + B_N(r)) * ldb
+                                                + B_K2(r));
+    }
+#pragma unroll
+    for (int r = 0; r < 2; ++r) {
+        lds[(A_R2(r) + 0) * LSTR + A_C(r)] = pa[r].x;
+        lds[(A_R2(r) + 1) * LSTR + A_C(r)] = pa[r].y;
+        *reinterpret_cast<f64x2*>(&lds[BM * LSTR + B_N(r) * LSTR + B_K2(r)])
+            = pb[r];
+    }
+
+    for (int64_t kt = 0; kt < ktiles; ++kt) {
+        __syncthreads();   // buf[kt&1] complete; everyone left iter kt-1
+        const double* As_c = lds + (kt & 1) * 2 * BM * LSTR;
+        const double* Bs_c = As_c + BM * LSTR;
+        if (kt + 1 < ktiles) {
+            const int64_t k0 = (kt + 1) * BK;
+#pragma unroll
+            for (int r = 0; r < 2; ++r) {
+                pa[r] = *reinterpret_cast<const f64x2*>(
+                    A + (k0 + A_C(r)) * lda + bm + A_R2(r));
+                pb[r] = *reinterpret_cast<const f64x2*>(
+                    B + (bn + B_N(r)) * ldb + k0 + B_K2(r));
+            }
+        }
+#pragma unroll
+        for (int kk = 0; kk < 4; ++kk) {
+            double a[4], b[2];
+            const int kof = kk * 4 + l4;
+#pragma unroll
+            for (int i = 0; i < 4; ++i)
+                a[i] = As_c[(wr + i * 16 + l16) * LSTR + kof];
+#pragma unroll
+            for (int j = 0; j < 2; ++j)
+                b[j] = Bs_c[(wc + j * 16 + l16) * LSTR + kof];
+#pragma unroll
+            for (int i = 0; i < 4; ++i)
+#pragma unroll
+                for (int j = 0; j < 2; ++j)
+                    acc[i][j] = __builtin_amdgcn_mfma_f64_16x16x4f64(
+                        a[i], b[j], acc[i][j], 0, 0, 0);
+        }
+        if (kt + 1 < ktiles) {
+            double* As_n = lds + ((kt + 1) & 1) * 2 * BM * LSTR;
+            double* Bs_n = As_n + BM * LSTR;
+#pragma unroll
+            for (int r = 0; r < 2; ++r) {
+                As_n[(A_R2(r) + 0) * LSTR + A_C(r)] = pa[r].x;
+                As_n[(A_R2(r) + 1) * LSTR + A_C(r)] = pa[r].y;
+                *reinterpret_cast<f64x2*>(&Bs_n[B_N(r) * LSTR + B_K2(r)])
+                    = pb[r];
+            }
+        }
+    }
+#undef A_C
+#undef A_R2
+#undef B_N
+#undef B_K2
+
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+#pragma unroll
+        for (int j = 0; j < 2; ++j) {
+            int64_t col = bn + wc + j * 16 + l16;
+            double* cp = C + col * ldc + bm + wr + i * 16 + l4;
+#pragma unroll
+            for (int qq = 0; qq < 4; ++qq) {
+                double v = alpha * acc[i][j][qq];
+                cp[4 * qq] = (beta == 0.0) ? v : v + beta * cp[4 * qq];
+            }
+        }
+    }
+}
+
 // Naive fallback for arbitrary shapes (small parity chunks).
 __global__ void gemm_f64_naive(const double* __restrict__ A,
                                const double* __restrict__ B,
@@ -445,6 +567,9 @@ int launch_gemm_f64(void* Cv, const void* Av, const void* Bv,
                                A, B, C, m, n, k, lda, ldb, ldc, alpha, beta);
         else if (variant == 3)
             hipLaunchKernelGGL(gemm_f64_mfma_v3, g, dim3(512), 0, s,
+                               A, B, C, m, n, k, lda, ldb, ldc, alpha, beta);
+        else if (variant == 5)
+            hipLaunchKernelGGL(gemm_f64_mfma_v5, g, dim3(512), 0, s,
                                A, B, C, m, n, k, lda, ldb, ldc, alpha, beta);
         else if (bk == 32 && k % 32 == 0)
             hipLaunchKernelGGL(gemm_f64_mfma_v2<32>, g, dim3(256), 0, s,
