@@ -387,3 +387,26 @@ def test_map2_scalar(dja):
         assert np.array_equal(out.localpart(), xi + 7)
         out.close(); di.close()
     d.close()
+
+
+# ------------------------------------------------------- large-n stress
+def test_large_chunk_sum_2e31(dja):
+    """>2^31 elements in one chunk: exercises 64-bit indexing paths
+    (17 GiB in HBM; 288 GB per GPU)."""
+    n = (1 << 31) + 5
+    d = dja.DArray((n,), "f64")
+    d.fill_(1.0)
+    assert dja.dsum(d) == float(n)
+    dja.scale_(d, 0.5)
+    assert dja.dsum(d) == n * 0.5
+    d.close()
+
+
+def test_large_f32_abs2_mapreduce(dja):
+    """cfg-5 shape on one GPU: 2^31 f32 mapreduce(abs2, +)."""
+    n = 1 << 31
+    d = dja.DArray((n,), "f32")
+    d.fill_(0.5)
+    got = dja.mapreduce("abs2", "add", d)
+    assert abs(got - n * 0.25) / (n * 0.25) < 1e-3
+    d.close()
