@@ -410,3 +410,32 @@ def test_large_f32_abs2_mapreduce(dja):
     got = dja.mapreduce("abs2", "add", d)
     assert abs(got - n * 0.25) / (n * 0.25) < 1e-3
     d.close()
+
+
+# ---------------------------------------------------------- error paths
+def test_error_paths(dja):
+    from distributedarrays_jl_amd import DArrayError
+    a = dja.drand((100,), "f64")
+    b = dja.drand((200,), "f64")
+    with pytest.raises(DArrayError):
+        dja.map_("sin", a, b)           # mismatched dims
+    with pytest.raises(DArrayError):
+        dja.dmap("sin", dja.drand((10,), "i64"))   # op invalid for i64
+    with pytest.raises(Exception):
+        dja.dmatmul(a, b)               # not 2-D
+    c = dja.drand((50,), "f64")
+    c.close()
+    with pytest.raises(DArrayError):
+        dja.dsum(c)                     # use after close
+    a.close(); b.close()
+    dja.d_closeall()
+
+
+def test_copy_semantics(dja):
+    x = philox.fill_uniform_f64(5000, seed=71)
+    d = dja.distribute(x)
+    e = d.copy()
+    dja.scale_(d, 2.0)                  # copy is deep: e unchanged
+    assert np.array_equal(e.localpart(), x)
+    assert np.array_equal(d.localpart(), 2.0 * x)
+    d.close(); e.close()
