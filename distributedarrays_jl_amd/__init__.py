@@ -25,7 +25,7 @@ from .ops import (map_, dmap, map2_, elementwise, map2_scalar_, elementwise_scal
                   dextrema, dmean, ddot, dnorm, dmatmul, dreduce_dims,
                   dsum_dims, dprod_dims, dmaximum_dims, dminimum_dims,
                   dmean_dims, dmatvec, gather_box, map_general,
-                  broadcast_fma_general, dsort)
+                  broadcast_fma_general, dsort, dtranspose, ddiag_lmul, ddiag_rmul)
 
 __all__ = [
     "DArray", "DArrayError", "comm", "geometry", "plan", "spmd",
@@ -37,4 +37,5 @@ __all__ = [
     "dreduce_dims", "dsum_dims", "dprod_dims", "dmaximum_dims",
     "dminimum_dims", "dmean_dims", "dmatvec",
     "gather_box", "map_general", "broadcast_fma_general", "dsort",
+    "dtranspose", "ddiag_lmul", "ddiag_rmul",
 ]

@@ -347,6 +347,18 @@ int da_gemm_f64(void* C, const void* A, const void* B,
                            st().stream);
 }
 
+int da_transpose(void* dst, const void* src, uint64_t m, uint64_t n,
+                 int dtype) {
+    DA_REQUIRE_INIT();
+    return launch_transpose(dst, src, m, n, dtype, st().stream);
+}
+
+int da_diag_scale(void* a, uint64_t m, uint64_t n, const void* diag,
+                  int side, int dtype) {
+    DA_REQUIRE_INIT();
+    return launch_diag_scale(a, m, n, diag, side, dtype, st().stream);
+}
+
 /* debug-only: MFMA lane-map probe (not part of the public ABI) */
 int dbg_mfma_probe_f64(const void* A, const void* B, void* out_c,
                        void* out_raw) {

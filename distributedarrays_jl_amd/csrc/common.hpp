@@ -87,6 +87,10 @@ int launch_bcast_fma(void* d, const void* a, const void* b, double c,
                      uint64_t n, int dtype, hipStream_t s);
 int launch_map2_scalar(int opcode, void* dst, const void* src, double c,
                        int rev, uint64_t n, int dtype, hipStream_t s);
+int launch_transpose(void* dst, const void* src, uint64_t m, uint64_t n,
+                     int dtype, hipStream_t s);
+int launch_diag_scale(void* a, uint64_t m, uint64_t n, const void* diag,
+                      int side, int dtype, hipStream_t s);
 int launch_axpby(void* y, const void* x, double alpha, double beta,
                  uint64_t n, int dtype, hipStream_t s);
 int launch_add(void* dest, const void* src, double scale, uint64_t n,

@@ -472,6 +472,80 @@ int launch_map2_scalar(int opcode, void* dst, const void* src, double c,
     return 0;
 }
 
+
+// ---------------------------------------------------- transpose / diag
+// Local 2-D transpose (dst = src^T), LDS-tiled so both sides stay
+// coalesced — the per-chunk piece of copy(::Transpose{<:DArray})
+// (linalg.jl:10-17).
+template <typename T>
+__global__ void transpose_kernel(T* __restrict__ dst,
+                                 const T* __restrict__ src,
+                                 uint64_t m, uint64_t n) {
+    __shared__ T tile[32][33];
+    uint64_t bi = (uint64_t)blockIdx.x * 32;
+    uint64_t bj = (uint64_t)blockIdx.y * 32;
+    int tx = threadIdx.x, ty = threadIdx.y;   // 32 x 8
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+        uint64_t i = bi + tx, j = bj + ty + r * 8;
+        if (i < m && j < n) tile[ty + r * 8][tx] = src[i + j * m];
+    }
+    __syncthreads();
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+        uint64_t j = bj + tx, i = bi + ty + r * 8;   // dst is n x m
+        if (j < n && i < m) dst[j + i * n] = tile[tx][ty + r * 8];
+    }
+}
+
+int launch_transpose(void* dst, const void* src, uint64_t m, uint64_t n,
+                     int dtype, hipStream_t s) {
+    if (m == 0 || n == 0) return 0;
+    dim3 t(32, 8), g((m + 31) / 32, (n + 31) / 32);
+    switch (dtype) {
+    case DA_F64: hipLaunchKernelGGL(transpose_kernel<double>, g, t, 0, s,
+                    (double*)dst, (const double*)src, m, n); break;
+    case DA_F32: hipLaunchKernelGGL(transpose_kernel<float>, g, t, 0, s,
+                    (float*)dst, (const float*)src, m, n); break;
+    case DA_I64: hipLaunchKernelGGL(transpose_kernel<int64_t>, g, t, 0, s,
+                    (int64_t*)dst, (const int64_t*)src, m, n); break;
+    default: return set_err(-3, "da_transpose: bad dtype %d", dtype);
+    }
+    DA_CHECK_HIP(hipGetLastError());
+    return 0;
+}
+
+// Diagonal scaling: side=0 rows (lmul!(Diagonal(d), A), linalg.jl:169-177),
+// side=1 cols (rmul!(A, Diagonal(d)), linalg.jl:179-187).
+template <typename T>
+__global__ void diag_scale_kernel(T* __restrict__ a, uint64_t m, uint64_t n,
+                                  const T* __restrict__ diag, int side) {
+    uint64_t e = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    uint64_t total = m * n;
+    for (uint64_t t = e; t < total; t += stride) {
+        uint64_t i = t % m, j = t / m;
+        a[t] = a[t] * diag[side == 0 ? i : j];
+    }
+}
+
+int launch_diag_scale(void* a, uint64_t m, uint64_t n, const void* diag,
+                      int side, int dtype, hipStream_t s) {
+    if (m == 0 || n == 0) return 0;
+    int g = nblocks(m * n);
+    switch (dtype) {
+    case DA_F64: hipLaunchKernelGGL(diag_scale_kernel<double>, dim3(g),
+                    dim3(TPB), 0, s, (double*)a, m, n,
+                    (const double*)diag, side); break;
+    case DA_F32: hipLaunchKernelGGL(diag_scale_kernel<float>, dim3(g),
+                    dim3(TPB), 0, s, (float*)a, m, n,
+                    (const float*)diag, side); break;
+    default: return set_err(-3, "da_diag_scale: bad dtype %d", dtype);
+    }
+    DA_CHECK_HIP(hipGetLastError());
+    return 0;
+}
+
 // ------------------------------------------- fused broadcast & BLAS-1 like
 template <typename T, bool NT>
 __global__ void bcast_fma_kernel(T* __restrict__ d, const T* __restrict__ a,

@@ -814,3 +814,58 @@ def dsort(d, samples_per_rank=64):
         buf.free()
     out.close()
     return res
+
+
+# ------------------------------------------- transpose / Diagonal scaling
+def dtranspose(D):
+    """copy(transpose(D)) — linalg.jl:10-17: the result DArray's chunk
+    (I) gathers D[reverse(I)] (remote gather -> gather_box over xGMI)
+    and transposes locally (LDS-tiled kernel)."""
+    import numpy as np
+    if D.ndims != 2:
+        raise DArrayError("dtranspose: 2-D only")
+    out = DArray((D.dims[1], D.dims[0]), D.dtype)
+    boxes = [None] * D.nranks
+    for c, r in enumerate(out.ranks):
+        (rlo, rhi), (clo, chi) = out.idxs[c]
+        boxes[r] = ((clo, chi), (rlo, rhi))   # reversed into D's coords
+    buf, shape = gather_box(D, boxes)
+    if buf is not None and out.lnumel:
+        check(lib.da_transpose(out._ptr(), buf.p, shape[0], shape[1],
+                               DTYPES[D.dtype]))
+        check(lib.da_synchronize())
+    if buf is not None:
+        buf.free()
+    return out
+
+
+def _diag_scale(D, dvec, side):
+    import numpy as np
+    dvec = np.ascontiguousarray(np.asarray(dvec,
+                                dtype=NUMPY_DTYPES[D.dtype]))
+    dim = 0 if side == 0 else 1
+    if D.ndims != 2 or dvec.shape != (D.dims[dim],):
+        raise DArrayError("diag scale: need 2-D DArray and matching diag")
+    if D.lnumel:
+        lo, hi = D.lidx[dim]
+        sl = np.ascontiguousarray(dvec[lo:hi])
+        buf = _Buf(max(sl.nbytes, 1))
+        check(lib.da_h2d(buf.p, sl.ctypes.data_as(ctypes.c_void_p),
+                         sl.nbytes))
+        check(lib.da_diag_scale(D._ptr(), D.lshape[0], D.lshape[1],
+                                buf.p, side, DTYPES[D.dtype]))
+        check(lib.da_synchronize())
+        buf.free()
+    return D
+
+
+def ddiag_lmul(dvec, D):
+    """lmul!(Diagonal(d), DA) — linalg.jl:169-177 (row scaling; each
+    rank receives only its slice of d, the DestinationSerializer
+    pattern)."""
+    return _diag_scale(D, dvec, 0)
+
+
+def ddiag_rmul(D, dvec):
+    """rmul!(DA, Diagonal(d)) — linalg.jl:179-187 (column scaling)."""
+    return _diag_scale(D, dvec, 1)

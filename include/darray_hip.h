@@ -134,6 +134,12 @@ int da_gemm_f64(void* C, const void* A, const void* B,
                 int64_t lda, int64_t ldb, int64_t ldc,
                 double alpha, double beta);
 
+/* ---- transpose / Diagonal scaling (linalg.jl:1-17, :169-187) --------- */
+int da_transpose(void* dst, const void* src, uint64_t m, uint64_t n,
+                 int dtype);                 /* dst = src^T, LDS-tiled */
+int da_diag_scale(void* a, uint64_t m, uint64_t n, const void* diag,
+                  int side, int dtype);      /* side 0: rows (lmul!), 1: cols (rmul!) */
+
 /* ---- distributed samplesort building blocks (src/sort.jl:103-170) --- */
 int da_sort(void* chunk, uint64_t n, int dtype);       /* per-chunk radix sort */
 int da_lower_bound(const void* sorted, uint64_t n, int dtype,

@@ -140,3 +140,37 @@ def test_gpu_map_general_aligned_fastpath():
     dja.map_general("abs2", o, d)
     assert np.array_equal(o.localpart(), x * x)
     o.close(); d.close()
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("shape", [(64, 64), (50, 30), (33, 97), (1, 7)])
+def test_gpu_dtranspose(shape):
+    import distributedarrays_jl_amd as dja
+    dja.comm.init()
+    m, n = shape
+    x = np.asfortranarray(philox.fill_uniform_f64(m * n, 21)
+                          .reshape(shape, order="F"))
+    d = dja.distribute(x)
+    t = dja.dtranspose(d)
+    assert t.dims == (n, m)
+    assert np.array_equal(t.collect(), np.asfortranarray(x.T))
+    t.close(); d.close()
+
+
+@pytest.mark.gpu
+def test_gpu_diag_scale():
+    import distributedarrays_jl_amd as dja
+    dja.comm.init()
+    m, n = 40, 30
+    x = np.asfortranarray(philox.fill_uniform_f64(m * n, 22)
+                          .reshape(m, n, order="F"))
+    dl = philox.fill_uniform_f64(m, 23)
+    dr = philox.fill_uniform_f64(n, 24)
+    d = dja.distribute(x)
+    dja.ddiag_lmul(dl, d)
+    ref = dl[:, None] * x
+    assert np.array_equal(d.localpart(), ref)
+    dja.ddiag_rmul(d, dr)
+    ref = ref * dr[None, :]
+    assert np.array_equal(d.localpart(), ref)
+    d.close()
