@@ -10,11 +10,32 @@
 // a tree, within the 1e-6 contract of BASELINE.json (the reference's own
 // fold order is likewise unspecified: docs/src/index.md:208-236).
 #include "common.hpp"
+#include <stdlib.h>
 
 namespace da {
 
 constexpr int RTPB = 256;          // 4 waves per block
-constexpr int RMAXB = 2048;        // block partials (fits scratch)
+constexpr int RMAXB = 8192;        // block partials (fits scratch)
+
+static inline int reduce_grid(uint64_t want) {
+    static int cap = -1;
+    if (cap < 0) {
+        const char* e = getenv("DA_RBLOCKS");
+        cap = e ? atoi(e) : 2048;
+        if (cap < 1 || cap > RMAXB) cap = 2048;
+    }
+    if (want < 1) want = 1;
+    return (int)(want > (uint64_t)cap ? (uint64_t)cap : want);
+}
+
+static inline bool reduce_v4() {
+    static int v = -1;
+    if (v < 0) {
+        const char* e = getenv("DA_RV4");
+        v = e ? atoi(e) : 0;
+    }
+    return v == 1;
+}
 
 template <typename T> struct RedIdent {
     static __device__ __host__ T get(int redop) {
@@ -119,6 +140,31 @@ __global__ void reduce_stage1(int mapop, int redop, const T* __restrict__ src,
     if (threadIdx.x == 0) partials[blockIdx.x] = acc;
 }
 
+// 32 B/thread-iteration variant (two dwordx4 per lane): more memory-
+// level parallelism per wave — A/B-gated via DA_RV4.
+template <typename T>
+__global__ void reduce_stage1_v4(int mapop, int redop,
+                                 const T* __restrict__ src, uint64_t n,
+                                 T* __restrict__ partials) {
+    uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    T acc = RedIdent<T>::get(redop);
+    using V = T __attribute__((ext_vector_type(4)));
+    uint64_t nv = n / 4;
+    const V* sv = reinterpret_cast<const V*>(src);
+    for (uint64_t j = i; j < nv; j += stride) {
+        V v = sv[j];
+        acc = red_comb(redop, acc, mapf<T>(mapop, (T)v.x));
+        acc = red_comb(redop, acc, mapf<T>(mapop, (T)v.y));
+        acc = red_comb(redop, acc, mapf<T>(mapop, (T)v.z));
+        acc = red_comb(redop, acc, mapf<T>(mapop, (T)v.w));
+    }
+    for (uint64_t j = 4 * nv + i; j < n; j += stride)
+        acc = red_comb(redop, acc, mapf<T>(mapop, src[j]));
+    acc = block_reduce(redop, acc);
+    if (threadIdx.x == 0) partials[blockIdx.x] = acc;
+}
+
 template <typename T>
 __global__ void reduce_stage2(int redop, const T* __restrict__ partials,
                               int np, T* __restrict__ out) {
@@ -136,14 +182,19 @@ static int do_reduce(int mapop, int redop, const T* src, uint64_t n,
         *(T*)out_host = RedIdent<T>::get(redop);
         return 0;
     }
-    uint64_t want = (n / 2 + RTPB - 1) / RTPB;
-    int g = (int)(want < 1 ? 1 : (want > RMAXB ? RMAXB : want));
+    bool v4 = reduce_v4();
+    uint64_t want = (n / (v4 ? 4 : 2) + RTPB - 1) / RTPB;
+    int g = reduce_grid(want);
     int rc = ensure_partials((RMAXB + 1) * sizeof(T));
     if (rc) return rc;
     T* parts = (T*)st().partials;
     T* dout = parts + RMAXB;
-    hipLaunchKernelGGL(reduce_stage1<T>, dim3(g), dim3(RTPB), 0, s,
-                       mapop, redop, src, n, parts);
+    if (v4)
+        hipLaunchKernelGGL(reduce_stage1_v4<T>, dim3(g), dim3(RTPB), 0, s,
+                           mapop, redop, src, n, parts);
+    else
+        hipLaunchKernelGGL(reduce_stage1<T>, dim3(g), dim3(RTPB), 0, s,
+                           mapop, redop, src, n, parts);
     DA_CHECK_HIP(hipGetLastError());
     hipLaunchKernelGGL(reduce_stage2<T>, dim3(1), dim3(RTPB), 0, s,
                        redop, parts, g, dout);
