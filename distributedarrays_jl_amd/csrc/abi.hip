@@ -366,6 +366,11 @@ int dbg_mfma_probe_f64(const void* A, const void* B, void* out_c,
     return dbg_mfma_probe_impl(A, B, out_c, out_raw, st().stream);
 }
 
+int dbg_mfma_probe_f32(const void* A, const void* B, void* out_raw) {
+    DA_REQUIRE_INIT();
+    return dbg_mfma_probe_f32_impl(A, B, out_raw, st().stream);
+}
+
 /* ---- point-to-point --------------------------------------------------- */
 
 int da_group_start(void) {

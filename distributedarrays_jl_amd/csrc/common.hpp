@@ -107,5 +107,7 @@ int launch_gemm_f64(void* C, const void* A, const void* B,
                     double alpha, double beta, hipStream_t s);
 int dbg_mfma_probe_impl(const void* A, const void* B, void* out_c,
                         void* out_raw, hipStream_t s);
+int dbg_mfma_probe_f32_impl(const void* A, const void* B, void* out_raw,
+                            hipStream_t s);
 
 } // namespace da

@@ -478,6 +478,32 @@ void gemm_f64_mfma_v5(const double* __restrict__ A,
     }
 }
 
+
+typedef float f32x4v __attribute__((ext_vector_type(4)));
+
+__global__ void mfma_probe_f32_kernel(const float* __restrict__ A,
+                                      const float* __restrict__ B,
+                                      float* __restrict__ out_raw) {
+    int l = threadIdx.x;
+    float a = A[(l & 15) + 16 * (l >> 4)];   // A[row, k] col-major 16x4
+    float b = B[(l >> 4) + 4 * (l & 15)];    // B[k, col] col-major 4x16
+    f32x4v acc = {0.f, 0.f, 0.f, 0.f};
+    acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, b, acc, 0, 0, 0);
+#pragma unroll
+    for (int q = 0; q < 4; ++q) out_raw[l * 4 + q] = acc[q];
+}
+
+extern "C" int dbg_mfma_probe_f32(const void* A, const void* B,
+                                  void* out_raw);
+int dbg_mfma_probe_f32_impl(const void* A, const void* B, void* out_raw,
+                            hipStream_t s) {
+    hipLaunchKernelGGL(mfma_probe_f32_kernel, dim3(1), dim3(64), 0, s,
+                       (const float*)A, (const float*)B, (float*)out_raw);
+    DA_CHECK_HIP(hipGetLastError());
+    DA_CHECK_HIP(hipStreamSynchronize(s));
+    return 0;
+}
+
 // Naive fallback for arbitrary shapes (small parity chunks).
 __global__ void gemm_f64_naive(const double* __restrict__ A,
                                const double* __restrict__ B,
