@@ -563,6 +563,165 @@ int dbg_mfma_probe_impl(const void* A, const void* B, void* out_c,
     return 0;
 }
 
+
+// f32 GEMM on the exact f32-input MFMA (v_mfma_f32_16x16x4_f32, 157 TF
+// peak = the f32 vector rate; bitwise an fmaf chain).  Same v3
+// structure; C/D lane map for f32 16x16x4 probed on hardware
+// (tools/probe_f32.py): col = lane&15, row = 4*(lane>>4) + q — NOTE:
+// opposite q/lane roles vs the f64 instruction.
+__global__ __launch_bounds__(512, 4)
+void gemm_f32_mfma_v3(const float* __restrict__ A,
+                      const float* __restrict__ B,
+                      float* __restrict__ C, int64_t m, int64_t n,
+                      int64_t k, int64_t lda, int64_t ldb, int64_t ldc,
+                      float alpha, float beta) {
+    typedef float f32x4 __attribute__((ext_vector_type(4)));
+    __shared__ float As[BM * LSTR];
+    __shared__ float Bs[BN * LSTR];
+
+    const int tid = threadIdx.x;
+    const int wave = tid >> 6;
+    const int lane = tid & 63;
+    const int l16 = lane & 15;
+    const int l4 = lane >> 4;
+    const int wr = (wave >> 2) * 64;
+    const int wc = (wave & 3) * 32;
+
+    const int gx = gridDim.x, nwg = gridDim.x * gridDim.y;
+    int w = blockIdx.y * gx + blockIdx.x;
+    int q = nwg >> 3, rmd = nwg & 7, xcd = w & 7, idx = w >> 3;
+    int sw = (xcd < rmd ? xcd * (q + 1) : rmd * (q + 1) + (xcd - rmd) * q)
+             + idx;
+    const int64_t bm = (int64_t)(sw % gx) * BM;
+    const int64_t bn = (int64_t)(sw / gx) * BN;
+
+    f32x4 acc[4][2];
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+#pragma unroll
+        for (int j = 0; j < 2; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+    // staging: A/B tiles are 2048 floats; 512 threads x 1 float4 each
+    const int a_c = tid >> 5, a_r4 = (tid & 31) * 4;
+    const int b_n = tid >> 2, b_k4 = (tid & 3) * 4;
+
+    f32x4 pa, pb;
+    const int64_t ktiles = k / BK;
+    pa = *reinterpret_cast<const f32x4*>(A + (int64_t)a_c * lda + bm + a_r4);
+    pb = *reinterpret_cast<const f32x4*>(B + (bn + b_n) * ldb + b_k4);
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+        As[(a_r4 + r) * LSTR + a_c] = pa[r];
+        Bs[b_n * LSTR + b_k4 + r] = pb[r];
+    }
+
+    for (int64_t kt = 0; kt < ktiles; ++kt) {
+        __syncthreads();
+        if (kt + 1 < ktiles) {
+            const int64_t k0 = (kt + 1) * BK;
+            pa = *reinterpret_cast<const f32x4*>(
+                A + (k0 + a_c) * lda + bm + a_r4);
+            pb = *reinterpret_cast<const f32x4*>(
+                B + (bn + b_n) * ldb + k0 + b_k4);
+        }
+#pragma unroll
+        for (int kk = 0; kk < 4; ++kk) {
+            float a[4], b[2];
+            const int kof = kk * 4 + l4;
+#pragma unroll
+            for (int i = 0; i < 4; ++i)
+                a[i] = As[(wr + i * 16 + l16) * LSTR + kof];
+#pragma unroll
+            for (int j = 0; j < 2; ++j)
+                b[j] = Bs[(wc + j * 16 + l16) * LSTR + kof];
+#pragma unroll
+            for (int i = 0; i < 4; ++i)
+#pragma unroll
+                for (int j = 0; j < 2; ++j)
+                    acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                        a[i], b[j], acc[i][j], 0, 0, 0);
+        }
+        __syncthreads();
+        if (kt + 1 < ktiles) {
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                As[(a_r4 + r) * LSTR + a_c] = pa[r];
+                Bs[b_n * LSTR + b_k4 + r] = pb[r];
+            }
+        }
+    }
+
+    // f32 C/D map: row = 4*l4 + q (consecutive per lane -> one f32x4)
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+#pragma unroll
+        for (int j = 0; j < 2; ++j) {
+            int64_t col = bn + wc + j * 16 + l16;
+            float* cp = C + col * ldc + bm + wr + i * 16 + 4 * l4;
+#pragma unroll
+            for (int qq = 0; qq < 4; ++qq) {
+                float v = alpha * acc[i][j][qq];
+                cp[qq] = (beta == 0.f) ? v : v + beta * cp[qq];
+            }
+        }
+    }
+}
+
+template <typename T>
+__global__ void gemm_naive_t(const T* __restrict__ A,
+                             const T* __restrict__ B, T* __restrict__ C,
+                             int64_t m, int64_t n, int64_t k,
+                             int64_t lda, int64_t ldb, int64_t ldc,
+                             T alpha, T beta) {
+    int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int64_t j = (int64_t)blockIdx.y * blockDim.y + threadIdx.y;
+    if (i >= m || j >= n) return;
+    T s = (T)0;
+    for (int64_t kk = 0; kk < k; ++kk)
+        s += A[kk * lda + i] * B[j * ldb + kk];
+    T v = alpha * s;
+    C[j * ldc + i] = (beta == (T)0) ? v : v + beta * C[j * ldc + i];
+}
+
+template <typename T>
+__global__ void scale_c_t(T* __restrict__ C, int64_t m, int64_t n,
+                          int64_t ldc, T beta) {
+    int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int64_t j = (int64_t)blockIdx.y * blockDim.y + threadIdx.y;
+    if (i >= m || j >= n) return;
+    C[j * ldc + i] = (beta == (T)0) ? (T)0 : beta * C[j * ldc + i];
+}
+
+int launch_gemm_f32(void* Cv, const void* Av, const void* Bv,
+                    int64_t m, int64_t n, int64_t k,
+                    int64_t lda, int64_t ldb, int64_t ldc,
+                    double alpha, double beta, hipStream_t s) {
+    if (m < 0 || n < 0 || k < 0)
+        return set_err(-3, "da_gemm_f32: bad shape");
+    if (m == 0 || n == 0) return 0;
+    const float* A = (const float*)Av;
+    const float* B = (const float*)Bv;
+    float* C = (float*)Cv;
+    float al = (float)alpha, be = (float)beta;
+    if (k == 0 || al == 0.f) {
+        dim3 t(64, 4), g((m + 63) / 64, (n + 3) / 4);
+        hipLaunchKernelGGL(scale_c_t<float>, g, t, 0, s, C, m, n, ldc, be);
+        DA_CHECK_HIP(hipGetLastError());
+        return 0;
+    }
+    if (m % BM == 0 && n % BN == 0 && k % BK == 0) {
+        dim3 g(m / BM, n / BN);
+        hipLaunchKernelGGL(gemm_f32_mfma_v3, g, dim3(512), 0, s,
+                           A, B, C, m, n, k, lda, ldb, ldc, al, be);
+    } else {
+        dim3 t(64, 4), g((m + 63) / 64, (n + 3) / 4);
+        hipLaunchKernelGGL(gemm_naive_t<float>, g, t, 0, s,
+                           A, B, C, m, n, k, lda, ldb, ldc, al, be);
+    }
+    DA_CHECK_HIP(hipGetLastError());
+    return 0;
+}
+
 int launch_gemm_f64(void* Cv, const void* Av, const void* Bv,
                     int64_t m, int64_t n, int64_t k,
                     int64_t lda, int64_t ldb, int64_t ldc,

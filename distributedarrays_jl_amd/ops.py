@@ -211,20 +211,21 @@ def dmatmul(A, B, alpha=1.0):
     """C = alpha * A * B — the `*` wrapper (linalg.jl:266-273) plus
     _matmatmul! (linalg.jl:190-253), re-expressed per plan.py.  f64 only
     (the cfg-4 metric path); local GEMM is the MFMA kernel."""
-    if A.dtype != "f64" or B.dtype != "f64":
-        raise DArrayError("dmatmul: f64 only (metric path)")
+    if A.dtype not in ("f64", "f32") or B.dtype != A.dtype:
+        raise DArrayError("dmatmul: f64/f32 only (matching dtypes)")
     if A.ndims != 2 or B.ndims != 2 or A.dims[1] != B.dims[0]:
         raise DArrayError("dmatmul: shape mismatch %r x %r"
                           % (A.dims, B.dims))
+    gemm_fn = lib.da_gemm_f64 if A.dtype == "f64" else lib.da_gemm_f32
     m, kk = A.dims
     n = B.dims[1]
     I, J = A.dist
     K = plan.c_grid(A.dist, B.dist)[1]
-    C = DArray((m, n), "f64", (I, K))
+    C = DArray((m, n), A.dtype, (I, K))
     C.fill_(0.0)
     r = A.rank
     pos = plan.a_rank_pos(r, A.dist)
-    esz = 8
+    esz = DTYPE_SIZE[A.dtype]
 
     pieces = plan.bslab_plan(A.dist, A.cuts[1], B.dims, B.dist, B.idxs)
     my_sends = [p for p in pieces if p[0] == r and p[1] != r]
@@ -298,9 +299,8 @@ def dmatmul(A, B, alpha=1.0):
             clo, chi = ccols[k]
             nk = chi - clo
             pk = _Buf(mloc * nk * esz)
-            check(lib.da_gemm_f64(pk.p, A._ptr(), slab.at(clo * kloc * esz),
-                                  mloc, nk, kloc, mloc, kloc, mloc,
-                                  1.0, 0.0))
+            check(gemm_fn(pk.p, A._ptr(), slab.at(clo * kloc * esz),
+                          mloc, nk, kloc, mloc, kloc, mloc, 1.0, 0.0))
             partials.append(pk)
 
     # partial exchange + ordered accumulation (linalg.jl:243-251)
@@ -329,11 +329,11 @@ def dmatmul(A, B, alpha=1.0):
             srcrank = i + I * j
             if srcrank == r:
                 check(lib.da_add(C._ptr(), partials[myk].p, float(alpha),
-                                 C.lnumel, DTYPES["f64"]))
+                                 C.lnumel, DTYPES[A.dtype]))
             else:
                 buf = precv[(srcrank, myk)]
                 check(lib.da_add(C._ptr(), buf.p, float(alpha),
-                                 C.lnumel, DTYPES["f64"]))
+                                 C.lnumel, DTYPES[A.dtype]))
 
     check(lib.da_synchronize())
     for _, b in sendbufs:

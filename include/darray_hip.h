@@ -133,6 +133,12 @@ int da_gemm_f64(void* C, const void* A, const void* B,
                 int64_t m, int64_t n, int64_t k,
                 int64_t lda, int64_t ldb, int64_t ldc,
                 double alpha, double beta);
+/* f32 local GEMM on the exact f32-input MFMA (v_mfma_f32_16x16x4_f32;
+ * bitwise an fmaf chain — cdna_hip_programming.md §3) */
+int da_gemm_f32(void* C, const void* A, const void* B,
+                int64_t m, int64_t n, int64_t k,
+                int64_t lda, int64_t ldb, int64_t ldc,
+                double alpha, double beta);
 
 /* ---- transpose / Diagonal scaling (linalg.jl:1-17, :169-187) --------- */
 int da_transpose(void* dst, const void* src, uint64_t m, uint64_t n,

@@ -439,3 +439,21 @@ def test_copy_semantics(dja):
     assert np.array_equal(e.localpart(), x)
     assert np.array_equal(d.localpart(), 2.0 * x)
     d.close(); e.close()
+
+
+# -------------------------------------------------------------- f32 gemm
+def test_gemm_f32(dja):
+    """f32 matmul on the exact f32 MFMA (bitwise an fmaf chain)."""
+    for m, k, n in [(256, 256, 256), (60, 40, 30), (384, 128, 128)]:
+        A = np.asfortranarray(philox.fill_uniform_f32(m * k, 46)
+                              .reshape(m, k, order="F"))
+        B = np.asfortranarray(philox.fill_uniform_f32(k * n, 47)
+                              .reshape(k, n, order="F"))
+        dA, dB = dja.distribute(A), dja.distribute(B)
+        C = dja.dmatmul(dA, dB)
+        assert C.dtype == "f32"
+        ref = A.astype(np.float64) @ B.astype(np.float64)
+        got = C.localpart().astype(np.float64)
+        err = np.abs(got - ref).max() / np.abs(ref).max()
+        assert err < 1e-5, (m, k, n, err)
+        C.close(); dA.close(); dB.close()
