@@ -22,7 +22,7 @@ from .darray import (DArray, dzeros, dones, dfill, drand, drandn,
                      bytes_in_use)
 from .ops import (map_, dmap, map2_, elementwise, map2_scalar_, elementwise_scalar, broadcast_fma, axpy_,
                   add_, scale_, mapreduce, dsum, dprod, dmaximum, dminimum,
-                  dextrema, dmean, ddot, dnorm, dmatmul, dreduce_dims,
+                  dextrema, dmean, dcount, dall, dany, ddot, dnorm, dmatmul, dreduce_dims,
                   dsum_dims, dprod_dims, dmaximum_dims, dminimum_dims,
                   dmean_dims, dmatvec, gather_box, map_general,
                   broadcast_fma_general, dsort, dtranspose, ddiag_lmul, ddiag_rmul)
@@ -33,7 +33,7 @@ __all__ = [
     "localpart", "localindices", "d_closeall", "bytes_in_use",
     "map_", "dmap", "map2_", "elementwise", "map2_scalar_", "elementwise_scalar", "broadcast_fma", "axpy_",
     "add_", "scale_", "mapreduce", "dsum", "dprod", "dmaximum",
-    "dminimum", "dextrema", "dmean", "ddot", "dnorm", "dmatmul",
+    "dminimum", "dextrema", "dmean", "dcount", "dall", "dany", "ddot", "dnorm", "dmatmul",
     "dreduce_dims", "dsum_dims", "dprod_dims", "dmaximum_dims",
     "dminimum_dims", "dmean_dims", "dmatvec",
     "gather_box", "map_general", "broadcast_fma_general", "dsort",

@@ -69,6 +69,8 @@ _sigs = {
                     i32),
     "da_gemm_f32": ([ptr, ptr, ptr, i64, i64, i64, i64, i64, i64, f64, f64],
                     i32),
+    "da_gemm_i64": ([ptr, ptr, ptr, i64, i64, i64, i64, i64, i64, i64, i64],
+                    i32),
     "da_group_start": ([], i32),
     "da_group_end": ([], i32),
     "da_send": ([ptr, u64, i32], i32),

@@ -23,7 +23,8 @@ MAP2_OPS = [
 MAP2_OP = {name: i for i, name in enumerate(MAP2_OPS)}
 
 RED_OPS = {"add": 0, "mul": 1, "min": 2, "max": 3}
-RED_FS = {"identity": 0, "abs": 1, "abs2": 2}
+RED_FS = {"identity": 0, "abs": 1, "abs2": 2, "isnan": 3, "isfinite": 4,
+          "nonzero": 5}
 RAND_KINDS = {"uniform": 0, "normal": 1}
 
 I64_MAP_OPS = {"identity", "neg", "abs", "abs2", "sign"}

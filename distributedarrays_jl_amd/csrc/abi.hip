@@ -356,6 +356,15 @@ int da_gemm_f32(void* C, const void* A, const void* B,
                            st().stream);
 }
 
+int da_gemm_i64(void* C, const void* A, const void* B,
+                int64_t m, int64_t n, int64_t k,
+                int64_t lda, int64_t ldb, int64_t ldc,
+                int64_t alpha, int64_t beta) {
+    DA_REQUIRE_INIT();
+    return launch_gemm_i64(C, A, B, m, n, k, lda, ldb, ldc, alpha, beta,
+                           st().stream);
+}
+
 int da_transpose(void* dst, const void* src, uint64_t m, uint64_t n,
                  int dtype) {
     DA_REQUIRE_INIT();

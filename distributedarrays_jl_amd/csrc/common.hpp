@@ -105,6 +105,10 @@ int launch_gemm_f64(void* C, const void* A, const void* B,
                     int64_t m, int64_t n, int64_t k,
                     int64_t lda, int64_t ldb, int64_t ldc,
                     double alpha, double beta, hipStream_t s);
+int launch_gemm_i64(void* C, const void* A, const void* B,
+                    int64_t m, int64_t n, int64_t k,
+                    int64_t lda, int64_t ldb, int64_t ldc,
+                    int64_t alpha, int64_t beta, hipStream_t s);
 int launch_gemm_f32(void* C, const void* A, const void* B,
                     int64_t m, int64_t n, int64_t k,
                     int64_t lda, int64_t ldb, int64_t ldc,

@@ -722,6 +722,19 @@ int launch_gemm_f32(void* Cv, const void* Av, const void* Bv,
     return 0;
 }
 
+int launch_gemm_i64(void* Cv, const void* Av, const void* Bv,
+                    int64_t m, int64_t n, int64_t k,
+                    int64_t lda, int64_t ldb, int64_t ldc,
+                    int64_t alpha, int64_t beta, hipStream_t s) {
+    if (m <= 0 || n <= 0) return 0;
+    dim3 t(64, 4), g((m + 63) / 64, (n + 3) / 4);
+    hipLaunchKernelGGL(gemm_naive_t<int64_t>, g, t, 0, s,
+                       (const int64_t*)Av, (const int64_t*)Bv,
+                       (int64_t*)Cv, m, n, k, lda, ldb, ldc, alpha, beta);
+    DA_CHECK_HIP(hipGetLastError());
+    return 0;
+}
+
 int launch_gemm_f64(void* Cv, const void* Av, const void* Bv,
                     int64_t m, int64_t n, int64_t k,
                     int64_t lda, int64_t ldb, int64_t ldc,

@@ -88,6 +88,9 @@ __device__ __forceinline__ T mapf(int mapop, T x) {
     case DA_REDF_IDENTITY: return x;
     case DA_REDF_ABS: return x < (T)0 ? (T)(-x) : x;
     case DA_REDF_ABS2: return x * x;
+    case DA_REDF_ISNAN: return (T)(x != x ? 1 : 0);
+    case DA_REDF_ISFINITE: return (T)(isfinite((double)x) ? 1 : 0);
+    case DA_REDF_NONZERO: return (T)(x != (T)0 ? 1 : 0);
     }
     return x;
 }
@@ -96,6 +99,9 @@ __device__ __forceinline__ int64_t mapf(int mapop, int64_t x) {
     case DA_REDF_IDENTITY: return x;
     case DA_REDF_ABS: return x < 0 ? (int64_t)(0ull - (uint64_t)x) : x;
     case DA_REDF_ABS2: return (int64_t)((uint64_t)x * (uint64_t)x);
+    case DA_REDF_ISNAN: return 0;
+    case DA_REDF_ISFINITE: return 1;
+    case DA_REDF_NONZERO: return x != 0 ? 1 : 0;
     }
     return x;
 }
@@ -367,7 +373,7 @@ static int do_reduce_dims(int mapop, int redop, const T* src,
 int launch_reduce_dims(int mapop, int redop, const void* src,
                        uint64_t inner, uint64_t axis, uint64_t outer,
                        int dtype, void* dst, hipStream_t s) {
-    if (mapop < 0 || mapop > DA_REDF_ABS2 || redop < 0 || redop > DA_RED_MAX)
+    if (mapop < 0 || mapop > DA_REDF_NONZERO || redop < 0 || redop > DA_RED_MAX)
         return set_err(-3, "da_reduce_dims: bad op (%d,%d)", mapop, redop);
     switch (dtype) {
     case DA_F64: return do_reduce_dims<double>(mapop, redop,
@@ -382,7 +388,7 @@ int launch_reduce_dims(int mapop, int redop, const void* src,
 
 int launch_reduce(int mapop, int redop, const void* src, uint64_t n,
                   int dtype, void* out_host, hipStream_t s) {
-    if (mapop < 0 || mapop > DA_REDF_ABS2 || redop < 0 || redop > DA_RED_MAX)
+    if (mapop < 0 || mapop > DA_REDF_NONZERO || redop < 0 || redop > DA_RED_MAX)
         return set_err(-3, "da_reduce: bad op (%d,%d)", mapop, redop);
     switch (dtype) {
     case DA_F64: return do_reduce<double>(mapop, redop, (const double*)src,

@@ -164,6 +164,26 @@ def dmean(d):
     return dsum(d) / d.size
 
 
+def dcount(pred, d):
+    """count(f, A) — mapreduce.jl:115-122 (pred in {nonzero, isnan,
+    isfinite})."""
+    return int(mapreduce(pred, "add", d))
+
+
+def dall(pred, d):
+    """all(f, A) — mapreduce.jl:97-104."""
+    if d.size == 0:
+        return True
+    return mapreduce(pred, "min", d) == 1
+
+
+def dany(pred, d):
+    """any(f, A) — mapreduce.jl:106-113."""
+    if d.size == 0:
+        return False
+    return mapreduce(pred, "max", d) == 1
+
+
 def ddot(x, y):
     """dot — linalg.jl:36-45 (aligned cuts: local fused path)."""
     _aligned(x, y)
@@ -211,12 +231,19 @@ def dmatmul(A, B, alpha=1.0):
     """C = alpha * A * B — the `*` wrapper (linalg.jl:266-273) plus
     _matmatmul! (linalg.jl:190-253), re-expressed per plan.py.  f64 only
     (the cfg-4 metric path); local GEMM is the MFMA kernel."""
-    if A.dtype not in ("f64", "f32") or B.dtype != A.dtype:
-        raise DArrayError("dmatmul: f64/f32 only (matching dtypes)")
+    if A.dtype not in ("f64", "f32", "i64") or B.dtype != A.dtype:
+        raise DArrayError("dmatmul: f64/f32/i64 only (matching dtypes)")
     if A.ndims != 2 or B.ndims != 2 or A.dims[1] != B.dims[0]:
         raise DArrayError("dmatmul: shape mismatch %r x %r"
                           % (A.dims, B.dims))
-    gemm_fn = lib.da_gemm_f64 if A.dtype == "f64" else lib.da_gemm_f32
+    if A.dtype == "f64":
+        gemm_fn = lib.da_gemm_f64
+    elif A.dtype == "f32":
+        gemm_fn = lib.da_gemm_f32
+    else:
+        def gemm_fn(C_, A_, B_, m_, n_, k_, la, lb, lc, al, be):
+            return lib.da_gemm_i64(C_, A_, B_, m_, n_, k_, la, lb, lc,
+                                   int(al), int(be))
     m, kk = A.dims
     n = B.dims[1]
     I, J = A.dist

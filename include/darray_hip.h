@@ -55,7 +55,11 @@ enum da_map2op {
 
 /* reduction: redop x mapop (src/mapreduce.jl:17-39, :97-131) */
 enum da_redop  { DA_RED_ADD = 0, DA_RED_MUL, DA_RED_MIN, DA_RED_MAX };
-enum da_redf   { DA_REDF_IDENTITY = 0, DA_REDF_ABS, DA_REDF_ABS2 };
+enum da_redf   { DA_REDF_IDENTITY = 0, DA_REDF_ABS, DA_REDF_ABS2,
+                 DA_REDF_ISNAN, DA_REDF_ISFINITE, DA_REDF_NONZERO };
+/* predicate mapops feed the all/any/count specials of
+ * src/mapreduce.jl:97-131: count = (pred, ADD); any = (pred, MAX) == 1;
+ * all = (pred, MIN) == 1 */
 
 /* rand kinds (drand/drandn, src/darray.jl:502-532) */
 enum da_randkind { DA_RAND_UNIFORM = 0, DA_RAND_NORMAL = 1 };
@@ -133,6 +137,11 @@ int da_gemm_f64(void* C, const void* A, const void* B,
                 int64_t m, int64_t n, int64_t k,
                 int64_t lda, int64_t ldb, int64_t ldc,
                 double alpha, double beta);
+/* i64 local GEMM (exact, wrap mod 2^64; Julia Int matmul parity) */
+int da_gemm_i64(void* C, const void* A, const void* B,
+                int64_t m, int64_t n, int64_t k,
+                int64_t lda, int64_t ldb, int64_t ldc,
+                int64_t alpha, int64_t beta);
 /* f32 local GEMM on the exact f32-input MFMA (v_mfma_f32_16x16x4_f32;
  * bitwise an fmaf chain — cdna_hip_programming.md §3) */
 int da_gemm_f32(void* C, const void* A, const void* B,

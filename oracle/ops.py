@@ -92,6 +92,14 @@ MAPRED_FS = {
     "identity": lambda x: x,
     "abs": np.abs,
     "abs2": lambda x: x * x,
+    # predicate maps feeding all/any/count (mapreduce.jl:97-131):
+    # 0/1-valued; count = (pred, add), any = (pred, max)==1,
+    # all = (pred, min)==1
+    "isnan": lambda x: (np.isnan(x) if x.dtype.kind == "f"
+                        else np.zeros_like(x)).astype(x.dtype),
+    "isfinite": lambda x: (np.isfinite(x) if x.dtype.kind == "f"
+                           else np.ones_like(x)).astype(x.dtype),
+    "nonzero": lambda x: (x != 0).astype(x.dtype),
 }
 RED_OPS = {
     "add": (np.add, lambda dt: dt.type(0)),

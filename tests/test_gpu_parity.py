@@ -457,3 +457,34 @@ def test_gemm_f32(dja):
         err = np.abs(got - ref).max() / np.abs(ref).max()
         assert err < 1e-5, (m, k, n, err)
         C.close(); dA.close(); dB.close()
+
+
+def test_count_all_any(dja):
+    x = philox.fill_uniform_f64(10000, seed=81)
+    x[5] = np.nan
+    x[77] = 0.0
+    d = dja.distribute(x)
+    assert dja.dcount("isnan", d) == 1
+    assert dja.dcount("nonzero", d) == 9999
+    assert not dja.dall("isfinite", d)
+    assert dja.dany("isnan", d)
+    d.close()
+    y = philox.fill_uniform_f64(1000, seed=82) + 0.5
+    dy = dja.distribute(y)
+    assert dja.dall("isfinite", dy)
+    assert not dja.dany("isnan", dy)
+    assert dja.dcount("nonzero", dy) == 1000
+    dy.close()
+
+
+def test_gemm_i64_exact(dja):
+    with np.errstate(over="ignore"):
+        m, k, n = 60, 50, 40
+        A = np.asfortranarray((philox.fill_int64(m * k, 48) % 1000)
+                              .reshape(m, k, order="F"))
+        B = np.asfortranarray((philox.fill_int64(k * n, 49) % 1000)
+                              .reshape(k, n, order="F"))
+        dA, dB = dja.distribute(A), dja.distribute(B)
+        C = dja.dmatmul(dA, dB)
+        assert np.array_equal(C.localpart(), A @ B)   # wrap-exact
+        C.close(); dA.close(); dB.close()
