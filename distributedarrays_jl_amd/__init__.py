@@ -19,7 +19,7 @@ from ._ffi import DArrayError
 from . import comm, geometry, plan, spmd
 from .darray import (DArray, dzeros, dones, dfill, drand, drandn,
                      distribute, localpart, localindices, d_closeall,
-                     bytes_in_use)
+                     bytes_in_use, ddata, dgather, locate)
 from .ops import (map_, dmap, map2_, elementwise, map2_scalar_, elementwise_scalar, broadcast_fma, axpy_,
                   add_, scale_, mapreduce, dsum, dprod, dmaximum, dminimum,
                   dextrema, dmean, dcount, dall, dany, ddot, dnorm, dmatmul, dreduce_dims,
@@ -31,6 +31,7 @@ __all__ = [
     "DArray", "DArrayError", "comm", "geometry", "plan", "spmd",
     "dzeros", "dones", "dfill", "drand", "drandn", "distribute",
     "localpart", "localindices", "d_closeall", "bytes_in_use",
+    "ddata", "dgather", "locate",
     "map_", "dmap", "map2_", "elementwise", "map2_scalar_", "elementwise_scalar", "broadcast_fma", "axpy_",
     "add_", "scale_", "mapreduce", "dsum", "dprod", "dmaximum",
     "dminimum", "dextrema", "dmean", "dcount", "dall", "dany", "ddot", "dnorm", "dmatmul",

@@ -314,6 +314,29 @@ def localindices(d):
     return d.localindices()
 
 
+def ddata(value, dtype="f64"):
+    """One value per rank (ddata, darray.jl:120-148): a DVector of
+    length nranks whose rank-r element is that rank's value."""
+    _auto_init()
+    rank, nr = comm.rank_info()
+    d = DArray((nr,), dtype, (nr,))
+    arr = np.asfortranarray(np.array([value],
+                                     dtype=np.dtype(NUMPY_DTYPES[dtype])))
+    d.set_localpart(arr)
+    return d
+
+
+def dgather(d):
+    """gather(d::DArray{T,1}) — darray.jl:150-157: one element per
+    chunk collected to every rank (control-plane gather)."""
+    return d.collect()
+
+
+def locate(d, *point):
+    """locate(d, I...) — darray.jl:448-456 (0-based chunk coords)."""
+    return geometry.locate(d.cuts, point)
+
+
 def d_closeall():
     """core.jl:98-103."""
     for d in list(_registry.values()):

@@ -488,3 +488,14 @@ def test_gemm_i64_exact(dja):
         C = dja.dmatmul(dA, dB)
         assert np.array_equal(C.localpart(), A @ B)   # wrap-exact
         C.close(); dA.close(); dB.close()
+
+
+def test_ddata_gather_locate(dja):
+    d = dja.ddata(42.5)
+    assert np.array_equal(d.collect(), np.array([42.5]))
+    assert np.array_equal(dja.dgather(d), np.array([42.5]))
+    d.close()
+    v = dja.drand((50,), "f64")
+    assert dja.locate(v, 0) == (0,)
+    assert dja.locate(v, 49) == (0,)
+    v.close()
