@@ -178,6 +178,34 @@ def main():
 
     D.close()
 
+    # ---------------- informational legs (N==1 only) --------------------
+    if world == 1 and not args.no_gemm:
+        gn = args.gemm_n
+        GA = dja.DArray((gn, gn), "f32"); GA.rand_()
+        GB = dja.DArray((gn, gn), "f32"); GB.rand_()
+        C = dja.dmatmul(GA, GB); C.close()
+        barrier()
+        t0 = time.perf_counter()
+        for _ in range(3):
+            C = dja.dmatmul(GA, GB); C.close()
+        barrier()
+        extra["gemm_f32_tflops"] = 2.0 * gn ** 3 * 3 / \
+            (time.perf_counter() - t0) / 1e12
+        GA.close(); GB.close()
+        log("[bench] gemm f32: %.1f TFLOP/s" % extra["gemm_f32_tflops"])
+
+        S = dja.drand((1 << 27,), "f64")
+        r = dja.dsort(S); r.close()
+        barrier()
+        t0 = time.perf_counter()
+        for _ in range(3):
+            r = dja.dsort(S); r.close()
+        barrier()
+        extra["sort_gkeys_per_s"] = (1 << 27) * 3 / \
+            (time.perf_counter() - t0) / 1e9
+        S.close()
+        log("[bench] sort: %.2f Gkeys/s" % extra["sort_gkeys_per_s"])
+
     # ---------------- CPU baseline (rank 0, N==1 only) ------------------
     cpu_baseline = None
     if rank == 0 and world == 1:
