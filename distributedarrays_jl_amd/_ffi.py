@@ -76,6 +76,10 @@ _sigs = {
     "da_send": ([ptr, u64, i32], i32),
     "da_recv": ([ptr, u64, i32], i32),
     "da_sendrecv": ([ptr, i32, ptr, i32, u64], i32),
+    "da_p2p_stream": ([i32], i32),
+    "da_comm_after_compute": ([], i32),
+    "da_main_after_comm": ([], i32),
+    "da_comm_sync": ([], i32),
     "da_bcast": ([ptr, u64, i32], i32),
     "da_barrier": ([], i32),
     "da_synchronize": ([], i32),

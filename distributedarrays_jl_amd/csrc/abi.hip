@@ -77,6 +77,12 @@ int da_init(int device, int rank, int nranks, const char* rccl_uid_path) {
         return set_err(-2, "da_init: bad rank %d/%d", rank, nranks);
     DA_CHECK_HIP(hipSetDevice(device));
     DA_CHECK_HIP(hipStreamCreateWithFlags(&st().stream, hipStreamNonBlocking));
+    DA_CHECK_HIP(hipStreamCreateWithFlags(&st().comm_stream,
+                                          hipStreamNonBlocking));
+    DA_CHECK_HIP(hipEventCreateWithFlags(&st().ev_main,
+                                         hipEventDisableTiming));
+    DA_CHECK_HIP(hipEventCreateWithFlags(&st().ev_comm,
+                                         hipEventDisableTiming));
     st().device = device;
     st().rank = rank;
     st().nranks = nranks;
@@ -127,7 +133,11 @@ int da_shutdown(void) {
         st().bytes_in_use = 0;
         (void)pool_trim_locked();
     }
-    if (st().stream) { hipStreamDestroy(st().stream); st().stream = nullptr; }
+    if (st().ev_main) { (void)hipEventDestroy(st().ev_main); st().ev_main = nullptr; }
+    if (st().ev_comm) { (void)hipEventDestroy(st().ev_comm); st().ev_comm = nullptr; }
+    if (st().comm_stream) { (void)hipStreamDestroy(st().comm_stream);
+                            st().comm_stream = nullptr; }
+    if (st().stream) { (void)hipStreamDestroy(st().stream); st().stream = nullptr; }
     st().inited = false;
     return 0;
 }
@@ -403,11 +413,15 @@ int da_group_end(void) {
     return 0;
 }
 
+static hipStream_t p2p_stream() {
+    return st().p2p_comm ? st().comm_stream : st().stream;
+}
+
 int da_send(const void* buf, uint64_t nbytes, int peer) {
     DA_REQUIRE_INIT();
     if (!st().comm) return set_err(-2, "da_send: no communicator");
     DA_CHECK_NCCL(ncclSend(buf, nbytes, ncclChar, peer, st().comm,
-                           st().stream));
+                           p2p_stream()));
     return 0;
 }
 
@@ -415,7 +429,35 @@ int da_recv(void* buf, uint64_t nbytes, int peer) {
     DA_REQUIRE_INIT();
     if (!st().comm) return set_err(-2, "da_recv: no communicator");
     DA_CHECK_NCCL(ncclRecv(buf, nbytes, ncclChar, peer, st().comm,
-                           st().stream));
+                           p2p_stream()));
+    return 0;
+}
+
+int da_p2p_stream(int use_comm) {
+    DA_REQUIRE_INIT();
+    st().p2p_comm = (use_comm != 0);
+    return 0;
+}
+
+int da_comm_after_compute(void) {
+    /* comm stream waits for everything queued so far on the main stream */
+    DA_REQUIRE_INIT();
+    DA_CHECK_HIP(hipEventRecord(st().ev_main, st().stream));
+    DA_CHECK_HIP(hipStreamWaitEvent(st().comm_stream, st().ev_main, 0));
+    return 0;
+}
+
+int da_main_after_comm(void) {
+    /* main stream waits for everything queued so far on the comm stream */
+    DA_REQUIRE_INIT();
+    DA_CHECK_HIP(hipEventRecord(st().ev_comm, st().comm_stream));
+    DA_CHECK_HIP(hipStreamWaitEvent(st().stream, st().ev_comm, 0));
+    return 0;
+}
+
+int da_comm_sync(void) {
+    DA_REQUIRE_INIT();
+    DA_CHECK_HIP(hipStreamSynchronize(st().comm_stream));
     return 0;
 }
 
@@ -425,9 +467,9 @@ int da_sendrecv(const void* sbuf, int peer_s, void* rbuf, int peer_r,
     if (!st().comm) return set_err(-2, "da_sendrecv: no communicator");
     DA_CHECK_NCCL(ncclGroupStart());
     DA_CHECK_NCCL(ncclSend(sbuf, nbytes, ncclChar, peer_s, st().comm,
-                           st().stream));
+                           p2p_stream()));
     DA_CHECK_NCCL(ncclRecv(rbuf, nbytes, ncclChar, peer_r, st().comm,
-                           st().stream));
+                           p2p_stream()));
     DA_CHECK_NCCL(ncclGroupEnd());
     return 0;
 }

@@ -21,6 +21,12 @@ struct State {
     int rank = 0;
     int nranks = 1;
     hipStream_t stream = nullptr;
+    // second stream for point-to-point overlap (matmul partial exchange
+    // rides xGMI while the next local GEMM runs); p2p ops route here
+    // while p2p_comm is set (da_p2p_stream)
+    hipStream_t comm_stream = nullptr;
+    bool p2p_comm = false;
+    hipEvent_t ev_main = nullptr, ev_comm = nullptr;
     ncclComm_t comm = nullptr;
     // small persistent device scratch for scalar allreduce / reduce outputs
     void* scratch = nullptr;

@@ -315,32 +315,52 @@ def dmatmul(A, B, alpha=1.0):
     elif my_sends or my_recvs or my_local:
         raise DArrayError("B owner outside A's grid unsupported (round 1)")
 
-    # local partial GEMMs: partial[k] = A_local @ slab[:, ccols[k]]
+    # local partial GEMMs fused with the per-k partial exchange
+    # (linalg.jl:218-251).  With DA_MM_OVERLAP (default on) each k's
+    # grouped send/recv rides the comm stream as soon as its GEMM is
+    # done, overlapping the xGMI transfer with GEMM k+1; every A-grid
+    # rank issues group k in ascending-k order, so the pairing argument
+    # of the single-group schedule carries over unchanged.
+    import os as _os
     ccols = geometry.ranges1d(C.cuts[1])
+    moves = plan.partial_plan(A.dist, K)
+    my_psends = [mv for mv in moves if mv[0] == r]
+    my_precvs = [mv for mv in moves if mv[1] == r]
+    overlap = (_os.environ.get("DA_MM_OVERLAP", "1") == "1"
+               and (my_psends or my_precvs))
+    precv = {}
+    for (src, dst, k) in my_precvs:
+        precv[(src, k)] = _Buf(max(C.lshape[0] * C.lshape[1], 1) * esz)
     partials = []
-    if pos is not None and A.lnumel:
-        i, j = pos
-        mloc = A.lshape[0]
-        kloc = A.lshape[1]
-        for k in range(K):
+    if overlap:
+        check(lib.da_p2p_stream(1))
+    for k in range(K):
+        if pos is not None and A.lnumel:
+            mloc = A.lshape[0]
+            kloc = A.lshape[1]
             clo, chi = ccols[k]
             nk = chi - clo
             pk = _Buf(mloc * nk * esz)
             check(gemm_fn(pk.p, A._ptr(), slab.at(clo * kloc * esz),
                           mloc, nk, kloc, mloc, kloc, mloc, 1.0, 0.0))
             partials.append(pk)
-
-    # partial exchange + ordered accumulation (linalg.jl:243-251)
-    moves = plan.partial_plan(A.dist, K)
-    my_psends = [mv for mv in moves if mv[0] == r]
-    my_precvs = [mv for mv in moves if mv[1] == r]
-    precv = {}
-    if my_psends or my_precvs:
-        for (src, dst, k) in my_precvs:
-            srci, srcj = src % I, src // I
-            mloc = C.lshape[0]
-            nk = C.lshape[1]
-            precv[(src, k)] = _Buf(mloc * nk * esz)
+        if overlap:
+            check(lib.da_comm_after_compute())
+            sends_k = [mv for mv in my_psends if mv[2] == k]
+            recvs_k = [mv for mv in my_precvs if mv[2] == k]
+            if sends_k or recvs_k:
+                check(lib.da_group_start())
+                for (src, dst, kk) in sends_k:
+                    nb = A.lshape[0] * (ccols[kk][1] - ccols[kk][0]) * esz
+                    check(lib.da_send(partials[kk].p, nb, dst))
+                for (src, dst, kk) in recvs_k:
+                    nb = C.lshape[0] * C.lshape[1] * esz
+                    check(lib.da_recv(precv[(src, kk)].p, nb, src))
+                check(lib.da_group_end())
+    if overlap:
+        check(lib.da_main_after_comm())
+        check(lib.da_p2p_stream(0))
+    elif my_psends or my_precvs:
         check(lib.da_group_start())
         for (src, dst, k) in my_psends:
             nb = A.lshape[0] * (ccols[k][1] - ccols[k][0]) * esz

@@ -169,6 +169,13 @@ int da_send(const void* buf, uint64_t nbytes, int peer);
 int da_recv(void* buf, uint64_t nbytes, int peer);
 int da_sendrecv(const void* sbuf, int peer_s, void* rbuf, int peer_r,
                 uint64_t nbytes);
+/* comm/compute overlap: route p2p ops onto a second stream and fence the
+ * two streams with events (the matmul partial exchange rides xGMI while
+ * the next local GEMM runs) */
+int da_p2p_stream(int use_comm);
+int da_comm_after_compute(void);
+int da_main_after_comm(void);
+int da_comm_sync(void);
 int da_bcast(void* buf, uint64_t nbytes, int root);
 int da_barrier(void);      /* device barrier over the communicator */
 

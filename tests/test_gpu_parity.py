@@ -499,3 +499,17 @@ def test_ddata_gather_locate(dja):
     assert dja.locate(v, 0) == (0,)
     assert dja.locate(v, 49) == (0,)
     v.close()
+
+
+def test_comm_stream_fences(dja):
+    """overlap-machinery ABI calls: stream routing + event fences are
+    valid on a single rank (the actual overlap runs at N>1)."""
+    from distributedarrays_jl_amd._ffi import lib, check
+    check(lib.da_p2p_stream(1))
+    check(lib.da_comm_after_compute())
+    d = dja.drand((1 << 16,), "f64")
+    check(lib.da_main_after_comm())
+    check(lib.da_comm_sync())
+    check(lib.da_p2p_stream(0))
+    assert dja.dsum(d) > 0
+    d.close()
