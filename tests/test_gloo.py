@@ -113,25 +113,33 @@ def _matmul_worker(rank, tmpfile, q, world=WORLD):
             else:
                 rq.wait()
 
-        # --- local partials + exchange (mirrors partial_plan) ---
+        # --- local partials + PER-K exchange rounds (the DA_MM_OVERLAP
+        # schedule of ops.dmatmul: every rank issues round k in
+        # ascending order; this validates that pairing with a real
+        # multiprocess transport) ---
         partials = [A_loc @ slab[:, ccols[k][0]:ccols[k][1]]
                     for k in range(K)]
         moves = plan.partial_plan(A_dist, K)
-        reqs = []
         got = {}
-        for tag, (src, dst, k) in enumerate(moves):
-            if src == rank:
-                t = torch.from_numpy(np.ascontiguousarray(partials[k]))
-                reqs.append((td.isend(t, dst, tag=1000 + tag), None, t))
-            elif dst == rank:
-                shp = pg.shape_of(C_idxs[rank])
-                t = torch.zeros(shp, dtype=torch.float64)
-                reqs.append((td.irecv(t, src, tag=1000 + tag),
-                             (src, k), t))
-        for rq, key, t in reqs:
-            rq.wait()
-            if key is not None:
-                got[key] = t.numpy()
+        for k in range(K):
+            reqs = []
+            for tag, (src, dst, kk) in enumerate(moves):
+                if kk != k:
+                    continue
+                if src == rank:
+                    t = torch.from_numpy(
+                        np.ascontiguousarray(partials[kk]))
+                    reqs.append((td.isend(t, dst, tag=1000 + tag),
+                                 None, t))
+                elif dst == rank:
+                    shp = pg.shape_of(C_idxs[rank])
+                    t = torch.zeros(shp, dtype=torch.float64)
+                    reqs.append((td.irecv(t, src, tag=1000 + tag),
+                                 (src, kk), t))
+            for rq, key, t in reqs:
+                rq.wait()
+                if key is not None:
+                    got[key] = t.numpy()
         C_loc = np.zeros(pg.shape_of(C_idxs[rank]), order="F")
         myk = rank // I
         for jj in plan.accumulate_order(J):
