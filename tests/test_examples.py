@@ -1,0 +1,16 @@
+"""The examples run and produce sane results (GPU)."""
+import pytest
+
+pytestmark = pytest.mark.gpu
+
+
+def test_monte_carlo_pi():
+    from examples.monte_carlo_pi import estimate_pi
+    pi = estimate_pi(1 << 22)
+    assert abs(pi - 3.14159265) < 0.01
+
+
+def test_power_iteration():
+    from examples.power_iteration import dominant_eig
+    lam = dominant_eig(512, iters=25)
+    assert abs(lam - 256) < 10   # uniform matrix: lambda ~ n/2
