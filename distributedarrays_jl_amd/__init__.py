@@ -19,24 +19,25 @@ from ._ffi import DArrayError
 from . import comm, geometry, plan, spmd
 from .darray import (DArray, dzeros, dones, dfill, drand, drandn,
                      distribute, localpart, localindices, d_closeall,
-                     bytes_in_use, ddata, dgather, locate)
+                     bytes_in_use, ddata, dgather, locate, allowscalar)
 from .ops import (map_, dmap, map2_, elementwise, map2_scalar_, elementwise_scalar, broadcast_fma, axpy_,
                   add_, scale_, mapreduce, dsum, dprod, dmaximum, dminimum,
                   dextrema, dmean, dcount, dall, dany, ddot, dnorm, dmatmul, dreduce_dims,
                   dsum_dims, dprod_dims, dmaximum_dims, dminimum_dims,
                   dmean_dims, dmatvec, gather_box, map_general,
-                  broadcast_fma_general, dsort, dtranspose, ddiag_lmul, ddiag_rmul)
+                  broadcast_fma_general, dsort, dtranspose, ddiag_lmul, ddiag_rmul,
+                  dgetindex, dmul_)
 
 __all__ = [
     "DArray", "DArrayError", "comm", "geometry", "plan", "spmd",
     "dzeros", "dones", "dfill", "drand", "drandn", "distribute",
     "localpart", "localindices", "d_closeall", "bytes_in_use",
-    "ddata", "dgather", "locate",
+    "ddata", "dgather", "locate", "allowscalar",
     "map_", "dmap", "map2_", "elementwise", "map2_scalar_", "elementwise_scalar", "broadcast_fma", "axpy_",
     "add_", "scale_", "mapreduce", "dsum", "dprod", "dmaximum",
     "dminimum", "dextrema", "dmean", "dcount", "dall", "dany", "ddot", "dnorm", "dmatmul",
     "dreduce_dims", "dsum_dims", "dprod_dims", "dmaximum_dims",
     "dminimum_dims", "dmean_dims", "dmatvec",
     "gather_box", "map_general", "broadcast_fma_general", "dsort",
-    "dtranspose", "ddiag_lmul", "ddiag_rmul",
+    "dtranspose", "ddiag_lmul", "ddiag_rmul", "dgetindex", "dmul_",
 ]

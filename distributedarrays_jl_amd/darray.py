@@ -26,6 +26,14 @@ import ctypes
 
 _registry = {}
 _next_id = [0]
+_allow_scalar = [True]
+
+
+def allowscalar(flag):
+    """allowscalar(flag) — darray.jl:638-640: gate element-wise scalar
+    indexing (the reference's tests disable it to force distributed
+    fast paths)."""
+    _allow_scalar[0] = bool(flag)
 
 
 def _auto_init():
@@ -244,6 +252,54 @@ class DArray:
             sl = tuple(slice(lo, hi) for lo, hi in self.idxs[c])
             out[sl] = gathered[self.ranks[c]]
         return out
+
+    def getindex(self, *point):
+        """Scalar D[i, j, ...] (darray.jl:645ff): owner reads one
+        element; result broadcast via the control plane at nranks>1.
+        Gated by allowscalar."""
+        if not _allow_scalar[0]:
+            raise _ffi.DArrayError(
+                "scalar indexing disallowed (allowscalar(False))")
+        sub = geometry.locate(self.cuts, point)
+        owner_chunk = geometry.grid_rank(sub, self.dist)
+        owner = self.ranks[owner_chunk]
+        val = None
+        if self.rank == owner:
+            off = 0
+            mul = 1
+            for dd in range(len(self.dims)):
+                off += (point[dd] - self.lidx[dd][0]) * mul
+                mul *= self.lshape[dd]
+            out = np.empty(1, dtype=np.dtype(NUMPY_DTYPES[self.dtype]))
+            check(lib.da_d2h(self.at_byte(off * DTYPE_SIZE[self.dtype]),
+                             out.ctypes.data_as(ctypes.c_void_p),
+                             DTYPE_SIZE[self.dtype]))
+            val = out[0]
+        if self.nranks > 1:
+            import torch.distributed as td
+            lst = [None] * self.nranks
+            td.all_gather_object(lst, val)
+            val = lst[owner]
+        return val
+
+    def setindex(self, value, *point):
+        """Scalar D[i, j, ...] = v (darray.jl:700ff); collective."""
+        if not _allow_scalar[0]:
+            raise _ffi.DArrayError(
+                "scalar indexing disallowed (allowscalar(False))")
+        sub = geometry.locate(self.cuts, point)
+        owner = self.ranks[geometry.grid_rank(sub, self.dist)]
+        if self.rank == owner:
+            off = 0
+            mul = 1
+            for dd in range(len(self.dims)):
+                off += (point[dd] - self.lidx[dd][0]) * mul
+                mul *= self.lshape[dd]
+            v = np.array([value], dtype=np.dtype(NUMPY_DTYPES[self.dtype]))
+            check(lib.da_h2d(self.at_byte(off * DTYPE_SIZE[self.dtype]),
+                             v.ctypes.data_as(ctypes.c_void_p),
+                             DTYPE_SIZE[self.dtype]))
+        return value
 
     def __eq__(self, other):
         if isinstance(other, DArray):

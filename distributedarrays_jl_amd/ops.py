@@ -227,6 +227,21 @@ def _copy2d(dst, dpitch, src, spitch, width, height):
         check(lib.da_copy2d(dst, dpitch, src, spitch, width, height))
 
 
+def dmul_(C, A, B, alpha=1.0, beta=0.0):
+    """mul!(C, A, B, alpha, beta) — full LinearAlgebra.mul! semantics
+    (linalg.jl:255): beta-scale C (fill 0 when beta==0,
+    linalg.jl:232-240) then accumulate alpha*A*B into it."""
+    R = dmatmul(A, B, alpha=alpha)
+    if beta == 0.0:
+        C.fill_(0.0)
+    elif beta != 1.0:
+        scale_(C, beta)
+    add_(C, R, 1.0)
+    check(lib.da_synchronize())
+    R.close()
+    return C
+
+
 def dmatmul(A, B, alpha=1.0):
     """C = alpha * A * B — the `*` wrapper (linalg.jl:266-273) plus
     _matmatmul! (linalg.jl:190-253), re-expressed per plan.py.  f64 only
@@ -709,6 +724,28 @@ def _dest_boxes(dest):
     for c, r in enumerate(dest.ranks):
         boxes[r] = dest.idxs[c]
     return boxes
+
+
+def dgetindex(A, *ranges):
+    """D[I...] for range indexing -> numpy array on every rank
+    (Array(view(A, I...)), the makelocal contract darray.jl:346-368;
+    collective: rank 0's box request is the union, gathered via
+    gather_box then broadcast on the control plane)."""
+    import numpy as np
+    box = tuple((lo, hi) for lo, hi in ranges)
+    boxes = [None] * A.nranks
+    boxes[A.rank] = box          # every rank requests the same box
+    for rr in range(A.nranks):
+        boxes[rr] = box
+    buf, shape = gather_box(A, boxes)
+    out = np.empty(shape, dtype=np.dtype(NUMPY_DTYPES[A.dtype]),
+                   order="F")
+    if buf is not None and out.size:
+        check(lib.da_d2h(buf.p, out.ctypes.data_as(ctypes.c_void_p),
+                         out.size * DTYPE_SIZE[A.dtype]))
+    if buf is not None:
+        buf.free()
+    return out
 
 
 def map_general(op, dest, src):

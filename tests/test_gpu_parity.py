@@ -513,3 +513,59 @@ def test_comm_stream_fences(dja):
     check(lib.da_p2p_stream(0))
     assert dja.dsum(d) > 0
     d.close()
+
+
+def test_scalar_indexing(dja):
+    from distributedarrays_jl_amd import allowscalar, DArrayError
+    x = np.asfortranarray(philox.fill_uniform_f64(40 * 30, 91)
+                          .reshape(40, 30, order="F"))
+    d = dja.distribute(x)
+    assert d.getindex(7, 11) == x[7, 11]
+    assert d.getindex(0, 0) == x[0, 0]
+    assert d.getindex(39, 29) == x[39, 29]
+    d.setindex(99.5, 5, 5)
+    assert d.getindex(5, 5) == 99.5
+    allowscalar(False)
+    with pytest.raises(DArrayError):
+        d.getindex(1, 1)
+    allowscalar(True)
+    d.close()
+
+
+def test_dgetindex_slices(dja):
+    x = np.asfortranarray(philox.fill_uniform_f64(50 * 40, 92)
+                          .reshape(50, 40, order="F"))
+    d = dja.distribute(x)
+    got = dja.dgetindex(d, (5, 45), (3, 33))
+    assert np.array_equal(got, x[5:45, 3:33])
+    d.close()
+
+
+def test_dmul_alpha_beta(dja):
+    m = 128
+    A = np.asfortranarray(philox.fill_uniform_f64(m * m, 93)
+                          .reshape(m, m, order="F"))
+    B = np.asfortranarray(philox.fill_uniform_f64(m * m, 94)
+                          .reshape(m, m, order="F"))
+    C0 = np.asfortranarray(philox.fill_uniform_f64(m * m, 95)
+                           .reshape(m, m, order="F"))
+    dA, dB, dC = dja.distribute(A), dja.distribute(B), dja.distribute(C0)
+    dja.dmul_(dC, dA, dB, alpha=2.0, beta=0.5)
+    ref = 2.0 * (A @ B) + 0.5 * C0
+    assert np.abs(dC.localpart() - ref).max() / np.abs(ref).max() < 1e-12
+    dA.close(); dB.close(); dC.close()
+
+
+def test_pool_soak(dja):
+    """alloc/free churn through the caching allocator with live ops."""
+    import distributedarrays_jl_amd as _dja
+    _dja.d_closeall()
+    base = _dja.bytes_in_use()
+    rng = np.random.default_rng(3)
+    for it in range(60):
+        n = int(rng.integers(1, 1 << 20))
+        d = _dja.drand((n,), "f64")
+        s = _dja.dsum(d)
+        assert 0 <= s <= n
+        d.close()
+    assert _dja.bytes_in_use() == base
