@@ -23,7 +23,7 @@ typedef double f64x2 __attribute__((ext_vector_type(2)));
 #define BM 128
 #define BN 128
 #define BK 16
-#define LSTR 18   // LDS row stride in doubles (bank-spread padding)
+#define LSTR 17   // LDS row stride (odd: staging writes 2-way not 4-way conflicted; frag reads stay conflict-free — profiles/r01 PMC)
 
 __global__ __launch_bounds__(256, 2)
 void gemm_f64_mfma(const double* __restrict__ A, const double* __restrict__ B,
@@ -73,7 +73,8 @@ void gemm_f64_mfma(const double* __restrict__ A, const double* __restrict__ B,
             int k2 = (idx & 7) * 2;           // k pair
             f64x2 v = *reinterpret_cast<const f64x2*>(
                 B + (bn + nn) * ldb + k0 + k2);
-            *reinterpret_cast<f64x2*>(&Bs[nn * LSTR + k2]) = v;
+            Bs[nn * LSTR + k2] = v.x;
+        Bs[nn * LSTR + k2 + 1] = v.y;
         }
         __syncthreads();
 #pragma unroll
@@ -179,7 +180,8 @@ void gemm_f64_mfma_v2(const double* __restrict__ A,
     for (int r = 0; r < NP; ++r) {
         As[(a_r2[r] + 0) * TSTR + a_c[r]] = pa[r].x;
         As[(a_r2[r] + 1) * TSTR + a_c[r]] = pa[r].y;
-        *reinterpret_cast<f64x2*>(&Bs[b_n[r] * TSTR + b_k2[r]]) = pb[r];
+        Bs[b_n[r] * TSTR + b_k2[r]] = pb[r].x;
+        Bs[b_n[r] * TSTR + b_k2[r] + 1] = pb[r].y;
     }
 
     for (int64_t kt = 0; kt < ktiles; ++kt) {
@@ -217,8 +219,8 @@ void gemm_f64_mfma_v2(const double* __restrict__ A,
             for (int r = 0; r < NP; ++r) {
                 As[(a_r2[r] + 0) * TSTR + a_c[r]] = pa[r].x;
                 As[(a_r2[r] + 1) * TSTR + a_c[r]] = pa[r].y;
-                *reinterpret_cast<f64x2*>(&Bs[b_n[r] * TSTR + b_k2[r]])
-                    = pb[r];
+                Bs[b_n[r] * TSTR + b_k2[r]] = pb[r].x;
+                Bs[b_n[r] * TSTR + b_k2[r] + 1] = pb[r].y;
             }
         }
     }
@@ -293,7 +295,8 @@ void gemm_f64_mfma_v3(const double* __restrict__ A,
     for (int r = 0; r < 2; ++r) {
         As[(A_R2(r) + 0) * LSTR + A_C(r)] = pa[r].x;
         As[(A_R2(r) + 1) * LSTR + A_C(r)] = pa[r].y;
-        *reinterpret_cast<f64x2*>(&Bs[B_N(r) * LSTR + B_K2(r)]) = pb[r];
+        Bs[B_N(r) * LSTR + B_K2(r)] = pb[r].x;
+        Bs[B_N(r) * LSTR + B_K2(r) + 1] = pb[r].y;
     }
 
     for (int64_t kt = 0; kt < ktiles; ++kt) {
@@ -331,8 +334,8 @@ void gemm_f64_mfma_v3(const double* __restrict__ A,
             for (int r = 0; r < 2; ++r) {
                 As[(A_R2(r) + 0) * LSTR + A_C(r)] = pa[r].x;
                 As[(A_R2(r) + 1) * LSTR + A_C(r)] = pa[r].y;
-                *reinterpret_cast<f64x2*>(&Bs[B_N(r) * LSTR + B_K2(r)])
-                    = pb[r];
+                Bs[B_N(r) * LSTR + B_K2(r)] = pb[r].x;
+                Bs[B_N(r) * LSTR + B_K2(r) + 1] = pb[r].y;
             }
         }
     }
@@ -411,8 +414,8 @@ void gemm_f64_mfma_v5(const double* __restrict__ A,
     for (int r = 0; r < 2; ++r) {
         lds[(A_R2(r) + 0) * LSTR + A_C(r)] = pa[r].x;
         lds[(A_R2(r) + 1) * LSTR + A_C(r)] = pa[r].y;
-        *reinterpret_cast<f64x2*>(&lds[BM * LSTR + B_N(r) * LSTR + B_K2(r)])
-            = pb[r];
+        lds[BM * LSTR + B_N(r) * LSTR + B_K2(r)] = pb[r].x;
+        lds[BM * LSTR + B_N(r) * LSTR + B_K2(r) + 1] = pb[r].y;
     }
 
     for (int64_t kt = 0; kt < ktiles; ++kt) {
@@ -453,8 +456,8 @@ void gemm_f64_mfma_v5(const double* __restrict__ A,
             for (int r = 0; r < 2; ++r) {
                 As_n[(A_R2(r) + 0) * LSTR + A_C(r)] = pa[r].x;
                 As_n[(A_R2(r) + 1) * LSTR + A_C(r)] = pa[r].y;
-                *reinterpret_cast<f64x2*>(&Bs_n[B_N(r) * LSTR + B_K2(r)])
-                    = pb[r];
+                Bs_n[B_N(r) * LSTR + B_K2(r)] = pb[r].x;
+                Bs_n[B_N(r) * LSTR + B_K2(r) + 1] = pb[r].y;
             }
         }
     }
