@@ -218,7 +218,10 @@ class DArray:
         return self
 
     def similar(self, dtype=None):
-        return DArray(self.dims, dtype or self.dtype, self.dist)
+        # propagate the chunk-owner mapping (non-identity for e.g.
+        # dims-reduction results) so elementwise ops stay aligned
+        return DArray(self.dims, dtype or self.dtype, self.dist,
+                      ranks=list(self.ranks))
 
     def copy(self):
         out = self.similar()
@@ -314,6 +317,9 @@ class DArray:
                 return all(flags)
             return same
         return NotImplemented
+
+    def __hash__(self):
+        return hash(self.id)
 
     def __repr__(self):
         return ("DArray(dims=%r, dist=%r, dtype=%s, rank=%d/%d, lshape=%r)"

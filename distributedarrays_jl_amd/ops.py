@@ -19,10 +19,12 @@ from .darray import DArray
 def _aligned(*ds):
     d0 = ds[0]
     for d in ds[1:]:
-        if d.dims != d0.dims or d.dist != d0.dist or d.dtype != d0.dtype:
+        if (d.dims != d0.dims or d.dist != d0.dist
+                or d.dtype != d0.dtype or d.ranks != d0.ranks):
             raise DArrayError(
-                "operands must share dims/dist/dtype (aligned cuts); "
-                "mismatched-cuts makelocal is a next-tier row, SURVEY §8f")
+                "operands must share dims/dist/dtype/owners (aligned "
+                "cuts); use map_general/broadcast_fma_general for "
+                "mismatched cuts")
     return d0
 
 

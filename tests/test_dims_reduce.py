@@ -136,3 +136,20 @@ def test_gpu_dims_reduce_large_axis_variant():
     assert np.allclose(R.collect(), x.sum(axis=1, keepdims=True),
                        rtol=1e-12)
     R.close(); d.close()
+
+
+@pytest.mark.gpu
+def test_ops_on_dims_reduction_result():
+    """similar()/map on a non-identity-ranks DArray (regression: similar
+    used to drop the owner mapping)."""
+    import distributedarrays_jl_amd as dja
+    dja.comm.init()
+    x = np.asfortranarray(philox.fill_uniform_f64(64 * 32, 17)
+                          .reshape(64, 32, order="F"))
+    d = dja.distribute(x)
+    R = dja.dsum_dims(d, (0,))
+    out = dja.dmap("abs2", R)
+    ref = x.sum(axis=0, keepdims=True) ** 2
+    assert np.allclose(out.collect(), ref, rtol=1e-12)
+    assert abs(dja.dsum(R) - x.sum()) < 1e-9
+    out.close(); R.close(); d.close()
