@@ -380,7 +380,7 @@ def ddata(value, dtype="f64"):
     """One value per rank (ddata, darray.jl:120-148): a DVector of
     length nranks whose rank-r element is that rank's value."""
     _auto_init()
-    rank, nr = comm.rank_info()
+    nr = comm.rank_info()[1]
     d = DArray((nr,), dtype, (nr,))
     arr = np.asfortranarray(np.array([value],
                                      dtype=np.dtype(NUMPY_DTYPES[dtype])))

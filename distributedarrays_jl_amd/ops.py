@@ -449,7 +449,6 @@ def dreduce_dims(f, op, d, dims):
     R = DArray(rdims, d.dtype, rdist, ranks=owners)
 
     me = d.rank
-    have_chunk = d.lchunk is not None and d.lnumel >= 0
     partial = None
     pshape = None
     if d.lchunk is not None:
