@@ -62,6 +62,7 @@ _sigs = {
     "da_transpose": ([ptr, ptr, u64, u64, i32], i32),
     "da_diag_scale": ([ptr, u64, u64, ptr, i32, i32], i32),
     "da_sort": ([ptr, u64, i32], i32),
+    "da_sort_out": ([ptr, ptr, u64, i32], i32),
     "da_lower_bound": ([ptr, u64, i32, ptr, i32,
                         ctypes.POINTER(ctypes.c_uint64)], i32),
     "da_allreduce": ([ptr, i32, i32, i32], i32),

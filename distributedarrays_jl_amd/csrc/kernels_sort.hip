@@ -11,32 +11,47 @@
 namespace da {
 
 template <typename T>
-static int do_sort(T* chunk, uint64_t n, hipStream_t s) {
-    if (n <= 1) return 0;
+static int do_sort_out(const T* src, T* dst, uint64_t n, hipStream_t s) {
+    if (n == 0) return 0;
+    if (n == 1) {
+        DA_CHECK_HIP(hipMemcpyAsync(dst, src, sizeof(T),
+                                    hipMemcpyDeviceToDevice, s));
+        return 0;
+    }
     size_t tmp_bytes = 0;
-    hipError_t e = rocprim::radix_sort_keys(nullptr, tmp_bytes, chunk,
+    hipError_t e = rocprim::radix_sort_keys(nullptr, tmp_bytes, src,
                                             (T*)nullptr, n, 0,
                                             sizeof(T) * 8, s);
     if (e != hipSuccess)
         return set_err(-(1000 + (int)e), "radix_sort query: %s",
                        hipGetErrorString(e));
-    void *out = nullptr, *tmp = nullptr;
-    int rc = da_alloc(n * sizeof(T), 0, &out);
+    void* tmp = nullptr;
+    int rc = da_alloc(tmp_bytes, 0, &tmp);
     if (rc) return rc;
-    rc = da_alloc(tmp_bytes, 0, &tmp);
-    if (rc) { da_free(out); return rc; }
-    e = rocprim::radix_sort_keys(tmp, tmp_bytes, chunk, (T*)out, n, 0,
+    e = rocprim::radix_sort_keys(tmp, tmp_bytes, src, dst, n, 0,
                                  sizeof(T) * 8, s);
     if (e != hipSuccess) {
-        da_free(out); da_free(tmp);
+        da_free(tmp);
         return set_err(-(1000 + (int)e), "radix_sort: %s",
                        hipGetErrorString(e));
     }
+    DA_CHECK_HIP(hipStreamSynchronize(s));
+    da_free(tmp);
+    return 0;
+}
+
+template <typename T>
+static int do_sort(T* chunk, uint64_t n, hipStream_t s) {
+    if (n <= 1) return 0;
+    void* out = nullptr;
+    int rc = da_alloc(n * sizeof(T), 0, &out);
+    if (rc) return rc;
+    rc = do_sort_out<T>((const T*)chunk, (T*)out, n, s);
+    if (rc) { da_free(out); return rc; }
     DA_CHECK_HIP(hipMemcpyAsync(chunk, out, n * sizeof(T),
                                 hipMemcpyDeviceToDevice, s));
     DA_CHECK_HIP(hipStreamSynchronize(s));
     da_free(out);
-    da_free(tmp);
     return 0;
 }
 
@@ -90,6 +105,22 @@ int da_sort(void* chunk, uint64_t n, int dtype) {
     case DA_I64: return do_sort<int64_t>((int64_t*)chunk, n, st().stream);
     }
     return set_err(-3, "da_sort: bad dtype %d", dtype);
+}
+
+/* Out-of-place ascending sort: src -> dst (saves the in-place variant's
+ * copy-back; the samplesort local stage sorts straight into the result
+ * chunk). */
+int da_sort_out(const void* src, void* dst, uint64_t n, int dtype) {
+    DA_REQUIRE_INIT();
+    switch (dtype) {
+    case DA_F64: return do_sort_out<double>((const double*)src,
+        (double*)dst, n, st().stream);
+    case DA_F32: return do_sort_out<float>((const float*)src,
+        (float*)dst, n, st().stream);
+    case DA_I64: return do_sort_out<int64_t>((const int64_t*)src,
+        (int64_t*)dst, n, st().stream);
+    }
+    return set_err(-3, "da_sort_out: bad dtype %d", dtype);
 }
 
 /* k lower-bound indices of host splitter values in a sorted device

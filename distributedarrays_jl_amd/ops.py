@@ -806,11 +806,15 @@ def dsort(d, samples_per_rank=64):
     esz = DTYPE_SIZE[d.dtype]
     npdt = np.dtype(NUMPY_DTYPES[d.dtype])
     P = d.nranks
+    if P == 1:
+        out = d.similar()
+        check(lib.da_sort_out(d._ptr(), out._ptr(), d.lnumel,
+                              DTYPES[d.dtype]))
+        check(lib.da_synchronize())
+        return out
     out = d.copy()
     if out.lnumel > 1:
         check(lib.da_sort(out._ptr(), out.lnumel, DTYPES[d.dtype]))
-    if P == 1:
-        return out
 
     import torch.distributed as td
     if not td.is_initialized():

@@ -157,6 +157,7 @@ int da_diag_scale(void* a, uint64_t m, uint64_t n, const void* diag,
 
 /* ---- distributed samplesort building blocks (src/sort.jl:103-170) --- */
 int da_sort(void* chunk, uint64_t n, int dtype);       /* per-chunk radix sort */
+int da_sort_out(const void* src, void* dst, uint64_t n, int dtype);
 int da_lower_bound(const void* sorted, uint64_t n, int dtype,
                    const void* splitters, int k, uint64_t* out);
 

@@ -23,6 +23,8 @@ EXACT_UNARY = {"neg", "abs", "abs2", "floor", "sign", "sqrt", "inv"}
 
 def _mk(dja, rng, pool, mirror):
     n = int(rng.integers(1, 20000))
+    if rng.integers(0, 10) == 0:       # occasional multi-MB array
+        n = int(rng.integers(1 << 20, 1 << 22))
     seed = int(rng.integers(0, 2 ** 31))
     d = dja.DArray((n,), "f64")
     d.rand_(seed_base=seed)

@@ -569,3 +569,24 @@ def test_pool_soak(dja):
         assert 0 <= s <= n
         d.close()
     assert _dja.bytes_in_use() == base
+
+
+def test_edge_values_reductions(dja):
+    """signed zeros, infinities, INT64 extremes."""
+    x = np.array([0.0, -0.0, 1.5, -np.inf, np.inf, 2.0, -3.5])
+    d = dja.distribute(np.ascontiguousarray(x))
+    assert dja.dmaximum(d) == np.inf
+    assert dja.dminimum(d) == -np.inf
+    assert dja.dsum(d) != dja.dsum(d) or np.isnan(x.sum()) == np.isnan(
+        dja.dsum(d))   # inf + -inf = nan on both sides
+    d.close()
+    with np.errstate(over="ignore"):
+        xi = np.array([np.iinfo(np.int64).min, np.iinfo(np.int64).max,
+                       -1, 0, 1], dtype=np.int64)
+        di = dja.distribute(xi)
+        assert dja.dmaximum(di) == np.iinfo(np.int64).max
+        assert dja.dminimum(di) == np.iinfo(np.int64).min
+        assert dja.dsum(di) == int(xi.sum())
+        out = dja.dmap("abs", di)   # abs(INT64_MIN) wraps like numpy
+        assert np.array_equal(out.localpart(), np.abs(xi))
+        out.close(); di.close()
