@@ -12,6 +12,11 @@ MAP_OPS = [
     "atanh", "sinpi", "cospi", "floor", "ceil",
     "round", "trunc", "sign", "deg2rad", "rad2deg",
     "sec", "csc", "cot",
+    "erf", "erfc", "erfinv", "erfcinv", "erfcx",
+    "gamma", "lgamma", "sinc", "cosc",
+    "sind", "cosd", "tand", "asind", "acosd",
+    "atand", "acot", "acotd", "asec", "acsc",
+    "asech", "acsch", "acoth",
 ]
 MAP_OP = {name: i for i, name in enumerate(MAP_OPS)}
 

@@ -220,6 +220,7 @@ template <> struct MathF<double> {
 template <typename T>
 __device__ __forceinline__ T apply_map(int op, T x) {
     const T one = (T)1, zero = (T)0;
+    (void)zero;
     switch (op) {
     case DA_OP_IDENTITY: return x;
     case DA_OP_NEG: return -x;
@@ -261,6 +262,31 @@ __device__ __forceinline__ T apply_map(int op, T x) {
     case DA_OP_SEC: return one / cos(x);
     case DA_OP_CSC: return one / sin(x);
     case DA_OP_COT: return one / tan(x);
+    case DA_OP_ERF: return erf(x);
+    case DA_OP_ERFC: return erfc(x);
+    case DA_OP_ERFINV: return erfinv(x);
+    case DA_OP_ERFCINV: return erfcinv(x);
+    case DA_OP_ERFCX: return erfcx(x);
+    case DA_OP_GAMMA: return tgamma(x);
+    case DA_OP_LGAMMA: return lgamma(x);
+    case DA_OP_SINC:   // Julia sinc: sin(pi x)/(pi x), 1 at 0
+        return x == zero ? one : sinpi(x) / ((T)M_PI * x);
+    case DA_OP_COSC:   // Julia cosc: d/dx sinc = cospi(x)/x - sinpi(x)/(pi x^2)
+        return x == zero ? zero
+                         : cospi(x) / x - sinpi(x) / ((T)M_PI * x * x);
+    case DA_OP_SIND: return sin(x * (T)(M_PI / 180.0));
+    case DA_OP_COSD: return cos(x * (T)(M_PI / 180.0));
+    case DA_OP_TAND: return tan(x * (T)(M_PI / 180.0));
+    case DA_OP_ASIND: return asin(x) * (T)(180.0 / M_PI);
+    case DA_OP_ACOSD: return acos(x) * (T)(180.0 / M_PI);
+    case DA_OP_ATAND: return atan(x) * (T)(180.0 / M_PI);
+    case DA_OP_ACOT: return atan(one / x);
+    case DA_OP_ACOTD: return atan(one / x) * (T)(180.0 / M_PI);
+    case DA_OP_ASEC: return acos(one / x);
+    case DA_OP_ACSC: return asin(one / x);
+    case DA_OP_ASECH: return acosh(one / x);
+    case DA_OP_ACSCH: return asinh(one / x);
+    case DA_OP_ACOTH: return atanh(one / x);
     }
     return x;
 }

@@ -42,6 +42,13 @@ enum da_mapop {
     DA_OP_ATANH, DA_OP_SINPI, DA_OP_COSPI, DA_OP_FLOOR, DA_OP_CEIL,
     DA_OP_ROUND, DA_OP_TRUNC, DA_OP_SIGN, DA_OP_DEG2RAD, DA_OP_RAD2DEG,
     DA_OP_SEC, DA_OP_CSC, DA_OP_COT,
+    /* the remainder of the reference's scalar-math list
+     * (test/darray.jl:775-800) that C math / OCML covers */
+    DA_OP_ERF, DA_OP_ERFC, DA_OP_ERFINV, DA_OP_ERFCINV, DA_OP_ERFCX,
+    DA_OP_GAMMA, DA_OP_LGAMMA, DA_OP_SINC, DA_OP_COSC,
+    DA_OP_SIND, DA_OP_COSD, DA_OP_TAND, DA_OP_ASIND, DA_OP_ACOSD,
+    DA_OP_ATAND, DA_OP_ACOT, DA_OP_ACOTD, DA_OP_ASEC, DA_OP_ACSC,
+    DA_OP_ASECH, DA_OP_ACSCH, DA_OP_ACOTH,
     DA_OP__N
 };
 

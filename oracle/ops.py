@@ -66,7 +66,38 @@ MAP_OPS = {
     "sec": lambda x: 1.0 / np.cos(x),
     "csc": lambda x: 1.0 / np.sin(x),
     "cot": lambda x: 1.0 / np.tan(x),
+    # special functions via scipy (the oracle is test infrastructure;
+    # the PRODUCT path uses OCML device builtins for these)
+    "erf": lambda x: _sp().erf(x),
+    "erfc": lambda x: _sp().erfc(x),
+    "erfinv": lambda x: _sp().erfinv(x),
+    "erfcinv": lambda x: _sp().erfcinv(x),
+    "erfcx": lambda x: _sp().erfcx(x),
+    "gamma": lambda x: _sp().gamma(x),
+    "lgamma": lambda x: _sp().gammaln(x),
+    "sinc": np.sinc,                      # numpy sinc IS Julia's sinc
+    "cosc": lambda x: np.where(
+        x == 0, 0.0, np.cos(np.pi * x) / np.where(x == 0, 1.0, x)
+        - np.sin(np.pi * x) / (np.pi * np.where(x == 0, 1.0, x) ** 2)),
+    "sind": lambda x: np.sin(np.deg2rad(x)),
+    "cosd": lambda x: np.cos(np.deg2rad(x)),
+    "tand": lambda x: np.tan(np.deg2rad(x)),
+    "asind": lambda x: np.rad2deg(np.arcsin(x)),
+    "acosd": lambda x: np.rad2deg(np.arccos(x)),
+    "atand": lambda x: np.rad2deg(np.arctan(x)),
+    "acot": lambda x: np.arctan(1.0 / x),
+    "acotd": lambda x: np.rad2deg(np.arctan(1.0 / x)),
+    "asec": lambda x: np.arccos(1.0 / x),
+    "acsc": lambda x: np.arcsin(1.0 / x),
+    "asech": lambda x: np.arccosh(1.0 / x),
+    "acsch": lambda x: np.arcsinh(1.0 / x),
+    "acoth": lambda x: np.arctanh(1.0 / x),
 }
+
+
+def _sp():
+    import scipy.special
+    return scipy.special
 
 # Binary elementwise ops (mapreduce.jl:180-189 specials + broadcast forms).
 # Integer div/mod/rem follow Julia: div = trunc, mod = floored, rem = trunc.

@@ -35,9 +35,12 @@ EXACT_OPS = ["identity", "neg", "abs", "abs2", "inv", "sqrt", "floor",
 TRANS_OPS = ["cbrt", "exp", "exp2", "exp10", "expm1", "log", "log2",
              "log10", "log1p", "sin", "cos", "tan", "asin", "acos",
              "atan", "sinh", "cosh", "tanh", "asinh", "atanh",
-             "sinpi", "cospi", "deg2rad", "rad2deg", "sec", "csc", "cot"]
+             "sinpi", "cospi", "deg2rad", "rad2deg", "sec", "csc", "cot",
+             "erf", "erfc", "erfinv", "erfcinv", "erfcx", "gamma",
+             "lgamma", "sinc", "cosc", "sind", "cosd", "tand", "asind",
+             "acosd", "atand", "acot", "acotd", "asech", "acsch"]
 # need |x| > 1 domain
-GT1_OPS = ["acosh"]
+GT1_OPS = ["acosh", "asec", "acsc", "acoth"]
 
 
 # ------------------------------------------------------------------ fills
@@ -86,8 +89,10 @@ def _input_for(op, n=100003):
         return x + 1.5
     if op in ("asin", "acos", "atanh"):
         return x * 0.99
-    if op in ("log", "log2", "log10", "sqrt", "inv", "csc", "cot"):
-        return x + 0.01
+    if op in ("log", "log2", "log10", "sqrt", "inv", "csc", "cot",
+              "gamma", "lgamma", "acot", "acotd", "acsch", "cosc",
+              "erfinv", "erfcinv", "asech"):
+        return x * 0.98 + 0.01
     return x
 
 
@@ -590,3 +595,16 @@ def test_edge_values_reductions(dja):
         out = dja.dmap("abs", di)   # abs(INT64_MIN) wraps like numpy
         assert np.array_equal(out.localpart(), np.abs(xi))
         out.close(); di.close()
+
+
+def test_sinc_cosc_special_points(dja):
+    x = np.array([0.0, 0.5, 1.0, -0.5, 2.0, -3.0])
+    d = dja.distribute(np.ascontiguousarray(x))
+    s = dja.dmap("sinc", d)
+    assert np.allclose(s.localpart(), np.sinc(x), rtol=1e-13, atol=1e-15)
+    assert s.localpart()[0] == 1.0
+    c = dja.dmap("cosc", d)
+    ref = oops.MAP_OPS["cosc"](x)
+    assert np.allclose(c.localpart(), ref, rtol=1e-12, atol=1e-14)
+    assert c.localpart()[0] == 0.0
+    s.close(); c.close(); d.close()
