@@ -112,7 +112,11 @@ def test_map_transcendental(dja, op):
     d = dja.distribute(x)
     out = dja.dmap(op, d)
     ref = oops.oracle_map(op, x)
-    assert np.allclose(out.localpart(), ref, rtol=1e-13, atol=1e-14), op
+    # cosc = cospi(x)/x - sinpi(x)/(pi x^2) cancels catastrophically at
+    # small x on BOTH sides (device vs libm roundings amplified); the
+    # 1e-6 contract still holds with margin
+    rtol, atol = (1e-9, 1e-9) if op == "cosc" else (1e-13, 1e-14)
+    assert np.allclose(out.localpart(), ref, rtol=rtol, atol=atol), op
     out.close(); d.close()
 
 
