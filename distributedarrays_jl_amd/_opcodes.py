@@ -17,6 +17,7 @@ MAP_OPS = [
     "sind", "cosd", "tand", "asind", "acosd",
     "atand", "acot", "acotd", "asec", "acsc",
     "asech", "acsch", "acoth",
+    "isnan", "isinf", "isfinite",
 ]
 MAP_OP = {name: i for i, name in enumerate(MAP_OPS)}
 

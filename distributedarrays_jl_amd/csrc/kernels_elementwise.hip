@@ -287,6 +287,9 @@ __device__ __forceinline__ T apply_map(int op, T x) {
     case DA_OP_ASECH: return acosh(one / x);
     case DA_OP_ACSCH: return asinh(one / x);
     case DA_OP_ACOTH: return atanh(one / x);
+    case DA_OP_ISNAN: return (T)(x != x ? 1 : 0);
+    case DA_OP_ISINF: return (T)(isinf((double)x) ? 1 : 0);
+    case DA_OP_ISFINITE: return (T)(isfinite((double)x) ? 1 : 0);
     }
     return x;
 }

@@ -49,6 +49,7 @@ enum da_mapop {
     DA_OP_SIND, DA_OP_COSD, DA_OP_TAND, DA_OP_ASIND, DA_OP_ACOSD,
     DA_OP_ATAND, DA_OP_ACOT, DA_OP_ACOTD, DA_OP_ASEC, DA_OP_ACSC,
     DA_OP_ASECH, DA_OP_ACSCH, DA_OP_ACOTH,
+    DA_OP_ISNAN, DA_OP_ISINF, DA_OP_ISFINITE,   /* 0/1-valued (Bool-array analog) */
     DA_OP__N
 };
 

@@ -92,6 +92,11 @@ MAP_OPS = {
     "asech": lambda x: np.arccosh(1.0 / x),
     "acsch": lambda x: np.arcsinh(1.0 / x),
     "acoth": lambda x: np.arctanh(1.0 / x),
+    # 0/1-valued predicates (the reference's Bool-array results, kept in
+    # the input dtype)
+    "isnan": lambda x: np.isnan(x).astype(x.dtype),
+    "isinf": lambda x: np.isinf(x).astype(x.dtype),
+    "isfinite": lambda x: np.isfinite(x).astype(x.dtype),
 }
 
 

@@ -612,3 +612,16 @@ def test_sinc_cosc_special_points(dja):
     assert np.allclose(c.localpart(), ref, rtol=1e-12, atol=1e-14)
     assert c.localpart()[0] == 0.0
     s.close(); c.close(); d.close()
+
+
+def test_predicate_maps(dja):
+    x = philox.fill_uniform_f64(1000, seed=99)
+    x[3] = np.nan
+    x[7] = np.inf
+    x[11] = -np.inf
+    d = dja.distribute(np.ascontiguousarray(x))
+    for op in ("isnan", "isinf", "isfinite"):
+        out = dja.dmap(op, d)
+        assert np.array_equal(out.localpart(), oops.MAP_OPS[op](x)), op
+        out.close()
+    d.close()
