@@ -41,24 +41,34 @@ def test_differential_fuzz():
     for _ in range(4):
         _mk(dja, rng, pool, mirror)
 
-    def checkeq(i, exact, tag):
+    def checkeq(i, tolv, tag):
         got = pool[i].localpart()
         ref = mirror[i]
         # long chains legitimately reach inf-inf = NaN on BOTH sides;
         # compare NaNs as equal
-        if exact:
+        if tolv is None:
             assert np.array_equal(got, ref, equal_nan=True), (tag, i)
-        else:
-            assert np.allclose(got, ref, rtol=1e-12, atol=1e-13,
-                               equal_nan=True), (tag, i)
+        elif tolv < 1e-6:
+            assert np.allclose(got, ref, rtol=tolv, atol=tolv,
+                               equal_nan=True), (tag, i, tolv)
 
-    exactness = [True] * len(pool)
+    # per-array relative error budget: None = bit-exact so far; a float
+    # is the tracked rtol bound, grown by each op's rough condition
+    # number; arrays past 1e-6 stay in the pool (path coverage) but are
+    # no longer value-compared.
+    exactness = [None] * len(pool)
+
+    def grow(tolv, x, factor=None):
+        if factor is None:
+            factor = 1.0 + float(np.nanmax(np.abs(x))) if x.size else 1.0
+        base = 1e-14 if tolv is None else tolv
+        return min(base * max(factor, 2.0) + 1e-14, 1.0)
     for step in range(nops):
         action = rng.integers(0, 8)
         i = int(rng.integers(0, len(pool)))
         if action == 0 and len(pool) < 10:
             _mk(dja, rng, pool, mirror)
-            exactness.append(True)
+            exactness.append(None)
         elif action == 1:  # unary map (maybe in-place)
             op = UNARY[int(rng.integers(0, len(UNARY)))]
             x = mirror[i]
@@ -67,16 +77,20 @@ def test_differential_fuzz():
                 dja.map_("abs", pool[i], pool[i])
                 mirror[i] = np.abs(x)
                 x = mirror[i]
+            ntol = (exactness[i] if op in EXACT_UNARY
+                    else grow(exactness[i], x))
+            if op in EXACT_UNARY and exactness[i] is not None:
+                ntol = grow(exactness[i], x, 2.0)   # exact op, inexact input
             if rng.integers(0, 2):
                 dja.map_(op, pool[i], pool[i])
                 mirror[i] = oops.MAP_OPS[op](x)
-                exactness[i] = exactness[i] and op in EXACT_UNARY
+                exactness[i] = ntol
                 checkeq(i, exactness[i], "map_" + op)
             else:
                 out = dja.dmap(op, pool[i])
                 pool.append(out)
                 mirror.append(oops.MAP_OPS[op](x))
-                exactness.append(exactness[i] and op in EXACT_UNARY)
+                exactness.append(ntol)
                 checkeq(len(pool) - 1, exactness[-1], "dmap_" + op)
         elif action == 2:  # binary with a same-shape partner (make one)
             op = BINARY[int(rng.integers(0, len(BINARY)))]
@@ -88,28 +102,34 @@ def test_differential_fuzz():
             out = dja.elementwise(op, pool[i], pool[j])
             pool.append(out)
             mirror.append(oops.MAP2_OPS[op](mirror[i], mirror[j]))
-            exactness.append(exactness[i] and exactness[j])
+            if exactness[i] is None and exactness[j] is None:
+                exactness.append(None)
+            else:
+                exactness.append(grow(max(exactness[i] or 1e-14,
+                                          exactness[j] or 1e-14),
+                                      mirror[-1], 4.0))
             checkeq(len(pool) - 1, exactness[-1], "bin_" + op)
         elif action == 3:  # scalar broadcast
             c = float(rng.uniform(-2, 2))
             out = dja.elementwise_scalar("add", pool[i], c)
             pool.append(out)
             mirror.append(mirror[i] + c)
-            exactness.append(exactness[i])
+            exactness.append(None if exactness[i] is None
+                             else grow(exactness[i], mirror[-1], 4.0))
             checkeq(len(pool) - 1, exactness[-1], "scalar_add")
         elif action == 4:  # reductions
             x = mirror[i]
+            tracked = exactness[i] is None or exactness[i] < 1e-8
             if np.isnan(x).any() or np.isinf(x).any():
-                assert not np.isfinite(dja.dsum(pool[i])) \
-                    or not exactness[i]
-            else:
+                dja.dsum(pool[i])   # path coverage only
+            elif tracked:
                 s = dja.dsum(pool[i])
                 ref = oops.oracle_reduce("identity", "add", [x])
-                tol = 1e-11 * max(1.0, abs(float(ref)))
-                assert abs(s - ref) <= tol or exactness[i] is False, "sum"
-                if x.size:
-                    assert (dja.dmaximum(pool[i]) == x.max()
-                            or not exactness[i])
+                tol = max(1e-11, (exactness[i] or 0) * 10) * \
+                    max(1.0, abs(float(ref)), float(np.abs(x).sum()))
+                assert abs(s - ref) <= tol, ("sum", i, s, ref)
+                if x.size and exactness[i] is None:
+                    assert dja.dmaximum(pool[i]) == x.max()
         elif action == 5:  # axpy / add / scale in place
             j = next((jj for jj in range(len(pool))
                       if jj != i and mirror[jj].shape == mirror[i].shape),
@@ -119,7 +139,12 @@ def test_differential_fuzz():
             a = float(rng.uniform(-1.5, 1.5))
             dja.axpy_(a, pool[j], pool[i])
             mirror[i] = mirror[i] + np.float64(a) * mirror[j]
-            exactness[i] = exactness[i] and exactness[j]
+            if exactness[i] is None and exactness[j] is None:
+                exactness[i] = None
+            else:
+                exactness[i] = grow(max(exactness[i] or 1e-14,
+                                        exactness[j] or 1e-14),
+                                    mirror[i], 4.0)
             checkeq(i, exactness[i], "axpy")
         elif action == 6 and len(pool) > 4:  # close + drop (pool churn)
             d = pool.pop(i)
@@ -130,8 +155,9 @@ def test_differential_fuzz():
             # differs from np.sort's NaN-last; skip NaN inputs)
             if mirror[i].size < 50000 and not np.isnan(mirror[i]).any():
                 r = dja.dsort(pool[i])
-                assert np.array_equal(r.localpart(), np.sort(mirror[i])) \
-                    or not exactness[i]
+                if exactness[i] is None:
+                    assert np.array_equal(r.localpart(),
+                                          np.sort(mirror[i]))
                 r.close()
     for d in pool:
         d.close()
