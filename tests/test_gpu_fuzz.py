@@ -63,6 +63,18 @@ def test_differential_fuzz():
             factor = 1.0 + float(np.nanmax(np.abs(x))) if x.size else 1.0
         base = 1e-14 if tolv is None else tolv
         return min(base * max(factor, 2.0) + 1e-14, 1.0)
+
+    def cancel_factor(a, b, r):
+        """amplification bound for a +/- b: (|a|+|b|) / |r|."""
+        if r.size == 0:
+            return 2.0
+        with np.errstate(invalid="ignore"):
+            num = float(np.nanmax(np.abs(a))) + float(np.nanmax(np.abs(b)))
+            rm = np.abs(r[np.isfinite(r) & (r != 0)])
+            rmin = float(rm.min()) if rm.size else 0.0
+        if rmin == 0.0 or not np.isfinite(num):
+            return 1e9          # full cancellation somewhere: untrack
+        return min(max(num / rmin, 2.0), 1e9)
     for step in range(nops):
         action = rng.integers(0, 8)
         i = int(rng.integers(0, len(pool)))
@@ -105,9 +117,11 @@ def test_differential_fuzz():
             if exactness[i] is None and exactness[j] is None:
                 exactness.append(None)
             else:
+                fac = (cancel_factor(mirror[i], mirror[j], mirror[-1])
+                       if op in ("add", "sub") else 4.0)
                 exactness.append(grow(max(exactness[i] or 1e-14,
                                           exactness[j] or 1e-14),
-                                      mirror[-1], 4.0))
+                                      mirror[-1], fac))
             checkeq(len(pool) - 1, exactness[-1], "bin_" + op)
         elif action == 3:  # scalar broadcast
             c = float(rng.uniform(-2, 2))
@@ -138,13 +152,15 @@ def test_differential_fuzz():
                 continue
             a = float(rng.uniform(-1.5, 1.5))
             dja.axpy_(a, pool[j], pool[i])
+            old_mirror_i = mirror[i]
             mirror[i] = mirror[i] + np.float64(a) * mirror[j]
             if exactness[i] is None and exactness[j] is None:
                 exactness[i] = None
             else:
+                fac = cancel_factor(old_mirror_i, mirror[j], mirror[i])
                 exactness[i] = grow(max(exactness[i] or 1e-14,
                                         exactness[j] or 1e-14),
-                                    mirror[i], 4.0)
+                                    mirror[i], fac)
             checkeq(i, exactness[i], "axpy")
         elif action == 6 and len(pool) > 4:  # close + drop (pool churn)
             d = pool.pop(i)
