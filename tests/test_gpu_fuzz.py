@@ -1,11 +1,16 @@
-"""Stateful differential fuzz: random sequences of hot-path ops applied
-to a pool of DArrays, mirrored step-by-step in numpy, compared after
-every op.  Exercises op interactions, the pooled allocator under churn,
-in-place aliasing, and mixed dtypes — the reference's differential
-pattern (DA vs Array on the same data) taken to random programs.
+"""Stateful differential fuzz: random op programs on a pool of DArrays,
+mirrored in numpy, with BIT-EXACT comparison along every chain of
+correctly-rounded ops (fills, +,-,*,min,max, sqrt, inv, abs, neg, abs2,
+floor, sign, scalar broadcast, axpy) — which is most of the op surface.
+Transcendental ops (exp/sin/...) make a chain 'unverified': its ops keep
+running (kernel/pool/aliasing coverage) but values are no longer
+compared, because error growth through long chains (cancellation,
+discontinuities, chaotic trig) is unbounded and directed tests already
+pin those ops at few-ulp tolerance on controlled inputs.
 
-Seeded and deterministic; FUZZ_OPS env scales the run (default 200 for
-the CI suite; tools/fuzz_long.py runs thousands)."""
+Catches: wrong buffers, aliasing, pooled-allocator reuse corruption,
+opcode mixups, chunk-size bookkeeping.  Seeded, deterministic; FUZZ_OPS
+scales the run (default 200)."""
 import os
 
 import numpy as np
@@ -15,13 +20,12 @@ from oracle import ops as oops, philox
 
 pytestmark = pytest.mark.gpu
 
-UNARY = ["neg", "abs", "abs2", "sqrt", "exp", "log1p", "sin", "cos",
-         "tanh", "floor", "sign", "inv"]
+EXACT_UNARY = ["neg", "abs", "abs2", "sqrt", "inv", "floor", "sign"]
+TRANS_UNARY = ["exp", "log1p", "sin", "cos", "tanh"]
 BINARY = ["add", "sub", "mul", "min2", "max2"]
-EXACT_UNARY = {"neg", "abs", "abs2", "floor", "sign", "sqrt", "inv"}
 
 
-def _mk(dja, rng, pool, mirror):
+def _mk(dja, rng, pool, mirror, exact):
     n = int(rng.integers(1, 20000))
     if rng.integers(0, 10) == 0:       # occasional multi-MB array
         n = int(rng.integers(1 << 20, 1 << 22))
@@ -30,6 +34,7 @@ def _mk(dja, rng, pool, mirror):
     d.rand_(seed_base=seed)
     pool.append(d)
     mirror.append(philox.fill_uniform_f64(n, seed))
+    exact.append(True)
 
 
 def test_differential_fuzz():
@@ -37,74 +42,47 @@ def test_differential_fuzz():
     dja.comm.init()
     nops = int(os.environ.get("FUZZ_OPS", "200"))
     rng = np.random.default_rng(20260915)
-    pool, mirror = [], []
+    pool, mirror, exact = [], [], []
     for _ in range(4):
-        _mk(dja, rng, pool, mirror)
+        _mk(dja, rng, pool, mirror, exact)
+    checked = [0]
 
-    def checkeq(i, tolv, tag):
+    def checkeq(i, tag):
+        if not exact[i]:
+            return
         got = pool[i].localpart()
-        ref = mirror[i]
-        # long chains legitimately reach inf-inf = NaN on BOTH sides;
-        # compare NaNs as equal
-        if tolv is None:
-            assert np.array_equal(got, ref, equal_nan=True), (tag, i)
-        elif tolv < 1e-6:
-            assert np.allclose(got, ref, rtol=tolv, atol=tolv,
-                               equal_nan=True), (tag, i, tolv)
+        assert np.array_equal(got, mirror[i], equal_nan=True), (tag, i)
+        checked[0] += 1
 
-    # per-array relative error budget: None = bit-exact so far; a float
-    # is the tracked rtol bound, grown by each op's rough condition
-    # number; arrays past 1e-6 stay in the pool (path coverage) but are
-    # no longer value-compared.
-    exactness = [None] * len(pool)
-
-    def grow(tolv, x, factor=None):
-        if factor is None:
-            factor = 1.0 + float(np.nanmax(np.abs(x))) if x.size else 1.0
-        base = 1e-14 if tolv is None else tolv
-        return min(base * max(factor, 2.0) + 1e-14, 1.0)
-
-    def cancel_factor(a, b, r):
-        """amplification bound for a +/- b: (|a|+|b|) / |r|."""
-        if r.size == 0:
-            return 2.0
-        with np.errstate(invalid="ignore"):
-            num = float(np.nanmax(np.abs(a))) + float(np.nanmax(np.abs(b)))
-            rm = np.abs(r[np.isfinite(r) & (r != 0)])
-            rmin = float(rm.min()) if rm.size else 0.0
-        if rmin == 0.0 or not np.isfinite(num):
-            return 1e9          # full cancellation somewhere: untrack
-        return min(max(num / rmin, 2.0), 1e9)
     for step in range(nops):
         action = rng.integers(0, 8)
         i = int(rng.integers(0, len(pool)))
         if action == 0 and len(pool) < 10:
-            _mk(dja, rng, pool, mirror)
-            exactness.append(None)
+            _mk(dja, rng, pool, mirror, exact)
         elif action == 1:  # unary map (maybe in-place)
-            op = UNARY[int(rng.integers(0, len(UNARY)))]
+            if rng.integers(0, 4) == 0:
+                op = TRANS_UNARY[int(rng.integers(0, len(TRANS_UNARY)))]
+                is_exact = False
+            else:
+                op = EXACT_UNARY[int(rng.integers(0, len(EXACT_UNARY)))]
+                is_exact = True
             x = mirror[i]
-            if op in ("sqrt", "log1p", "inv"):
-                # keep domain positive: abs first
+            if op in ("sqrt", "inv", "log1p"):
                 dja.map_("abs", pool[i], pool[i])
                 mirror[i] = np.abs(x)
                 x = mirror[i]
-            ntol = (exactness[i] if op in EXACT_UNARY
-                    else grow(exactness[i], x))
-            if op in EXACT_UNARY and exactness[i] is not None:
-                ntol = grow(exactness[i], x, 2.0)   # exact op, inexact input
             if rng.integers(0, 2):
                 dja.map_(op, pool[i], pool[i])
                 mirror[i] = oops.MAP_OPS[op](x)
-                exactness[i] = ntol
-                checkeq(i, exactness[i], "map_" + op)
+                exact[i] = exact[i] and is_exact
+                checkeq(i, "map_" + op)
             else:
                 out = dja.dmap(op, pool[i])
                 pool.append(out)
                 mirror.append(oops.MAP_OPS[op](x))
-                exactness.append(ntol)
-                checkeq(len(pool) - 1, exactness[-1], "dmap_" + op)
-        elif action == 2:  # binary with a same-shape partner (make one)
+                exact.append(exact[i] and is_exact)
+                checkeq(len(pool) - 1, "dmap_" + op)
+        elif action == 2:  # binary with a same-shape partner
             op = BINARY[int(rng.integers(0, len(BINARY)))]
             j = next((jj for jj in range(len(pool))
                       if jj != i and mirror[jj].shape == mirror[i].shape),
@@ -114,37 +92,26 @@ def test_differential_fuzz():
             out = dja.elementwise(op, pool[i], pool[j])
             pool.append(out)
             mirror.append(oops.MAP2_OPS[op](mirror[i], mirror[j]))
-            if exactness[i] is None and exactness[j] is None:
-                exactness.append(None)
-            else:
-                fac = (cancel_factor(mirror[i], mirror[j], mirror[-1])
-                       if op in ("add", "sub") else 4.0)
-                exactness.append(grow(max(exactness[i] or 1e-14,
-                                          exactness[j] or 1e-14),
-                                      mirror[-1], fac))
-            checkeq(len(pool) - 1, exactness[-1], "bin_" + op)
-        elif action == 3:  # scalar broadcast
+            exact.append(exact[i] and exact[j])
+            checkeq(len(pool) - 1, "bin_" + op)
+        elif action == 3:  # scalar broadcast (correctly rounded)
             c = float(rng.uniform(-2, 2))
             out = dja.elementwise_scalar("add", pool[i], c)
             pool.append(out)
             mirror.append(mirror[i] + c)
-            exactness.append(None if exactness[i] is None
-                             else grow(exactness[i], mirror[-1], 4.0))
-            checkeq(len(pool) - 1, exactness[-1], "scalar_add")
-        elif action == 4:  # reductions
+            exact.append(exact[i])
+            checkeq(len(pool) - 1, "scalar_add")
+        elif action == 4:  # reductions (verified on exact finite chains)
             x = mirror[i]
-            tracked = exactness[i] is None or exactness[i] < 1e-8
-            if np.isnan(x).any() or np.isinf(x).any():
-                dja.dsum(pool[i])   # path coverage only
-            elif tracked:
-                s = dja.dsum(pool[i])
+            s = dja.dsum(pool[i])
+            if exact[i] and x.size and np.isfinite(x).all():
                 ref = oops.oracle_reduce("identity", "add", [x])
-                tol = max(1e-11, (exactness[i] or 0) * 10) * \
-                    max(1.0, abs(float(ref)), float(np.abs(x).sum()))
+                tol = 1e-11 * max(1.0, float(np.abs(x).sum()))
                 assert abs(s - ref) <= tol, ("sum", i, s, ref)
-                if x.size and exactness[i] is None:
-                    assert dja.dmaximum(pool[i]) == x.max()
-        elif action == 5:  # axpy / add / scale in place
+                assert dja.dmaximum(pool[i]) == x.max()
+                checked[0] += 1
+        elif action == 5:  # axpy in place (two correctly-rounded ops,
+            # evaluated identically on both sides)
             j = next((jj for jj in range(len(pool))
                       if jj != i and mirror[jj].shape == mirror[i].shape),
                      None)
@@ -152,29 +119,23 @@ def test_differential_fuzz():
                 continue
             a = float(rng.uniform(-1.5, 1.5))
             dja.axpy_(a, pool[j], pool[i])
-            old_mirror_i = mirror[i]
             mirror[i] = mirror[i] + np.float64(a) * mirror[j]
-            if exactness[i] is None and exactness[j] is None:
-                exactness[i] = None
-            else:
-                fac = cancel_factor(old_mirror_i, mirror[j], mirror[i])
-                exactness[i] = grow(max(exactness[i] or 1e-14,
-                                        exactness[j] or 1e-14),
-                                    mirror[i], fac)
-            checkeq(i, exactness[i], "axpy")
+            exact[i] = exact[i] and exact[j]
+            checkeq(i, "axpy")
         elif action == 6 and len(pool) > 4:  # close + drop (pool churn)
             d = pool.pop(i)
             mirror.pop(i)
-            exactness.pop(i)
+            exact.pop(i)
             d.close()
-        elif action == 7:  # sort round trip (radix NaN-bit order
-            # differs from np.sort's NaN-last; skip NaN inputs)
-            if mirror[i].size < 50000 and not np.isnan(mirror[i]).any():
+        elif action == 7:  # sort round trip
+            if (mirror[i].size < 50000 and exact[i]
+                    and not np.isnan(mirror[i]).any()):
                 r = dja.dsort(pool[i])
-                if exactness[i] is None:
-                    assert np.array_equal(r.localpart(),
-                                          np.sort(mirror[i]))
+                assert np.array_equal(r.localpart(), np.sort(mirror[i]))
                 r.close()
+                checked[0] += 1
+    # the run must have actually verified a healthy number of ops
+    assert checked[0] >= nops // 10, checked[0]
     for d in pool:
         d.close()
     assert dja.bytes_in_use() == 0
