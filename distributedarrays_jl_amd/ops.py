@@ -640,12 +640,11 @@ def gather_box(A, boxes_all):
     into a device buffer (column-major).  Returns (_Buf, shape) or
     (None, None) when this rank requested nothing.  1-D/2-D only.
 
-    Fully-local requests degrade to one on-device strided copy (the
+    Fully-local requests degrade to on-device strided copies (the
     reference's zero-copy view case); remote pieces move as grouped
-    ncclSend/Recv of packed sub-blocks over xGMI."""
+    ncclSend/Recv of packed sub-blocks over xGMI.  N-D: dims >= 3
+    iterate per 2-D slice."""
     nd = A.ndims
-    if nd > 2:
-        raise DArrayError("gather_box: 1-D/2-D only (round 1)")
     esz = DTYPE_SIZE[A.dtype]
     mybox = boxes_all[A.rank] if A.rank < len(boxes_all) else None
     pieces = plan.halo_plan(A.idxs, A.ranks, boxes_all)
@@ -660,27 +659,56 @@ def gather_box(A, boxes_all):
         oshape = tuple(hi - lo for lo, hi in mybox)
         out = _Buf(max(geometry.nelems(mybox), 1) * esz)
 
-    def box2d(box):
-        if nd == 1:
-            return box[0], (0, 1)
-        return box[0], box[1]
-
     def nelems(box):
         n = 1
         for lo, hi in box:
             n *= hi - lo
         return n
 
+    def copy_box(dst_base, dst_shape, dst_org, src_base, src_shape,
+                 src_org, box):
+        """Copy an N-D box between two column-major buffers; dst/src_org
+        are the global coordinates of each buffer's origin.  Dims >= 3
+        iterate in Python (one hipMemcpy2DAsync per 2-D slice)."""
+        rows = box[0][1] - box[0][0]
+        cols = (box[1][1] - box[1][0]) if nd >= 2 else 1
+        def org_off(shape, org, outer_idx):
+            off = box[0][0] - org[0]
+            mul = shape[0]
+            if nd >= 2:
+                off += (box[1][0] - org[1]) * mul
+                mul *= shape[1]
+            for d in range(2, nd):
+                off += (box[d][0] + outer_idx[d - 2] - org[d]) * mul
+                mul *= shape[d]
+            return off
+        outer_dims = [box[d][1] - box[d][0] for d in range(2, nd)]
+        idx = [0] * len(outer_dims)
+        while True:
+            soff = org_off(src_shape, src_org, idx) * esz
+            doff = org_off(dst_shape, dst_org, idx) * esz
+            _copy2d(ctypes.c_void_p(dst_base.value + doff),
+                    dst_shape[0] * esz,
+                    ctypes.c_void_p(src_base.value + soff),
+                    src_shape[0] * esz, rows * esz, cols)
+            k = 0
+            while k < len(outer_dims):
+                idx[k] += 1
+                if idx[k] < outer_dims[k]:
+                    break
+                idx[k] = 0
+                k += 1
+            if k == len(outer_dims):
+                break
+
+    lorg = tuple(lo for lo, _ in A.lidx)
+
     sendbufs = []
     for (src, dst, box) in my_sends:
-        (rlo, rhi), (clo, chi) = box2d(box)
-        (llo, _), (lco, _) = box2d(A.lidx)
         buf = _Buf(nelems(box) * esz)
-        srows = A.lshape[0]
-        off = ((rlo - llo) + (clo - lco) * srows) * esz
-        _copy2d(buf.p, (rhi - rlo) * esz,
-                ctypes.c_void_p(A._ptr().value + off), srows * esz,
-                (rhi - rlo) * esz, chi - clo)
+        bshape = tuple(hi - lo for lo, hi in box)
+        borg = tuple(lo for lo, _ in box)
+        copy_box(buf.p, bshape, borg, A._ptr(), A.lshape, lorg, box)
         sendbufs.append((box, buf, dst))
     recvbufs = []
     for (src, dst, box) in my_recvs:
@@ -693,22 +721,13 @@ def gather_box(A, boxes_all):
             check(lib.da_recv(buf.p, nelems(box) * esz, src))
         check(lib.da_group_end())
     if mybox is not None:
-        (mlo, _), (mco, _) = box2d(mybox)
-        orows = oshape[0]
+        morg = tuple(lo for lo, _ in mybox)
         for (src, dst, box) in my_local:
-            (rlo, rhi), (clo, chi) = box2d(box)
-            (llo, _), (lco, _) = box2d(A.lidx)
-            srows = A.lshape[0]
-            soff = ((rlo - llo) + (clo - lco) * srows) * esz
-            doff = ((rlo - mlo) + (clo - mco) * orows) * esz
-            _copy2d(out.at(doff), orows * esz,
-                    ctypes.c_void_p(A._ptr().value + soff), srows * esz,
-                    (rhi - rlo) * esz, chi - clo)
+            copy_box(out.p, oshape, morg, A._ptr(), A.lshape, lorg, box)
         for box, buf, src in recvbufs:
-            (rlo, rhi), (clo, chi) = box2d(box)
-            doff = ((rlo - mlo) + (clo - mco) * orows) * esz
-            _copy2d(out.at(doff), orows * esz, buf.p, (rhi - rlo) * esz,
-                    (rhi - rlo) * esz, chi - clo)
+            bshape = tuple(hi - lo for lo, hi in box)
+            borg = tuple(lo for lo, _ in box)
+            copy_box(out.p, oshape, morg, buf.p, bshape, borg, box)
     check(lib.da_synchronize())
     for _, buf, _ in sendbufs:
         buf.free()

@@ -174,3 +174,24 @@ def test_gpu_diag_scale():
     ref = ref * dr[None, :]
     assert np.array_equal(d.localpart(), ref)
     d.close()
+
+
+@pytest.mark.gpu
+def test_gpu_gather_box_3d():
+    import distributedarrays_jl_amd as dja
+    dja.comm.init()
+    shape = (20, 16, 12)
+    x = np.asfortranarray(philox.fill_uniform_f64(int(np.prod(shape)), 31)
+                          .reshape(shape, order="F"))
+    d = dja.distribute(x)
+    box = ((3, 17), (2, 14), (1, 11))
+    buf, bshape = dja.gather_box(d, [box])
+    assert bshape == (14, 12, 10)
+    import ctypes
+    from distributedarrays_jl_amd._ffi import lib, check
+    out = np.empty(bshape, dtype=np.float64, order="F")
+    check(lib.da_d2h(buf.p, out.ctypes.data_as(ctypes.c_void_p),
+                     out.size * 8))
+    assert np.array_equal(out, x[3:17, 2:14, 1:11])
+    buf.free()
+    d.close()
