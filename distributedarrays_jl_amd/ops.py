@@ -638,7 +638,7 @@ def gather_box(A, boxes_all):
     SAME boxes_all list (boxes_all[r] = the box rank r requests, derived
     from shared metadata, or None) and receives its own box gathered
     into a device buffer (column-major).  Returns (_Buf, shape) or
-    (None, None) when this rank requested nothing.  1-D/2-D only.
+    (None, None) when this rank requested nothing.
 
     Fully-local requests degrade to on-device strided copies (the
     reference's zero-copy view case); remote pieces move as grouped
