@@ -204,6 +204,11 @@ class DArray:
 
     # ---- device-side content ops (darray.jl:822-834) ----
     def fill_(self, v):
+        if self.dtype == "i64" and abs(int(v)) > (1 << 53):
+            # the ABI carries the fill value as a double
+            raise _ffi.DArrayError(
+                "fill_: |i64 value| > 2^53 not representable through the "
+                "double fill parameter")
         if self.lnumel:
             check(lib.da_fill(self._ptr(), float(v), self.lnumel,
                               DTYPES[self.dtype]))

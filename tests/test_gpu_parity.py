@@ -625,3 +625,13 @@ def test_predicate_maps(dja):
         assert np.array_equal(out.localpart(), oops.MAP_OPS[op](x)), op
         out.close()
     d.close()
+
+
+def test_i64_fill_guard(dja):
+    from distributedarrays_jl_amd import DArrayError
+    d = dja.DArray((10,), "i64")
+    d.fill_(1 << 40)
+    assert (d.localpart() == (1 << 40)).all()
+    with pytest.raises(DArrayError):
+        d.fill_((1 << 60) + 1)
+    d.close()
