@@ -41,7 +41,7 @@ def test_differential_fuzz():
     import distributedarrays_jl_amd as dja
     dja.comm.init()
     nops = int(os.environ.get("FUZZ_OPS", "200"))
-    rng = np.random.default_rng(20260915)
+    rng = np.random.default_rng(int(os.environ.get("FUZZ_SEED", "20260915")))
     pool, mirror, exact = [], [], []
     for _ in range(4):
         _mk(dja, rng, pool, mirror, exact)
