@@ -37,11 +37,12 @@ def _mk(dja, rng, pool, mirror, exact):
     exact.append(True)
 
 
-def test_differential_fuzz():
+@pytest.mark.parametrize("seed", [20260915, 1, 2])
+def test_differential_fuzz(seed):
     import distributedarrays_jl_amd as dja
     dja.comm.init()
     nops = int(os.environ.get("FUZZ_OPS", "200"))
-    rng = np.random.default_rng(int(os.environ.get("FUZZ_SEED", "20260915")))
+    rng = np.random.default_rng(int(os.environ.get("FUZZ_SEED", seed)))
     pool, mirror, exact = [], [], []
     for _ in range(4):
         _mk(dja, rng, pool, mirror, exact)
@@ -50,8 +51,20 @@ def test_differential_fuzz():
     def checkeq(i, tag):
         if not exact[i]:
             return
-        got = pool[i].localpart()
-        assert np.array_equal(got, mirror[i], equal_nan=True), (tag, i)
+        ref = mirror[i]
+        if ref.size > (1 << 20):
+            # big arrays: bit-compare a 256 KiB head window (keeps long
+            # soaks cheap; heads catch wholesale buffer/opcode bugs)
+            import ctypes
+            k = 1 << 15
+            got = np.empty(k, dtype=np.float64)
+            from distributedarrays_jl_amd._ffi import lib as _l, check as _c
+            _c(_l.da_d2h(pool[i]._ptr(),
+                         got.ctypes.data_as(ctypes.c_void_p), k * 8))
+            assert np.array_equal(got, ref[:k], equal_nan=True), (tag, i)
+        else:
+            got = pool[i].localpart()
+            assert np.array_equal(got, ref, equal_nan=True), (tag, i)
         checked[0] += 1
 
     for step in range(nops):
