@@ -224,7 +224,7 @@ __device__ __forceinline__ T apply_map(int op, T x) {
     switch (op) {
     case DA_OP_IDENTITY: return x;
     case DA_OP_NEG: return -x;
-    case DA_OP_ABS: return x < zero ? -x : x;
+    case DA_OP_ABS: return fabs(x);   // fabs: abs(-0.0) = +0.0 (Julia/numpy)
     case DA_OP_ABS2: return x * x;
     case DA_OP_INV: return one / x;
     case DA_OP_SQRT: return sqrt(x);

@@ -86,7 +86,7 @@ template <typename T>
 __device__ __forceinline__ T mapf(int mapop, T x) {
     switch (mapop) {
     case DA_REDF_IDENTITY: return x;
-    case DA_REDF_ABS: return x < (T)0 ? (T)(-x) : x;
+    case DA_REDF_ABS: return fabs(x);   // abs(-0.0) = +0.0
     case DA_REDF_ABS2: return x * x;
     case DA_REDF_ISNAN: return (T)(x != x ? 1 : 0);
     case DA_REDF_ISFINITE: return (T)(isfinite((double)x) ? 1 : 0);

@@ -635,3 +635,19 @@ def test_i64_fill_guard(dja):
     with pytest.raises(DArrayError):
         d.fill_((1 << 60) + 1)
     d.close()
+
+
+def test_abs_signed_zero(dja):
+    """fuzz-found: abs(-0.0) must be +0.0 (Julia flipsign / numpy), so
+    1/abs(-0.0) is +inf not -inf."""
+    x = np.array([-0.0, 0.0, -1.5, 2.0])
+    d = dja.distribute(np.ascontiguousarray(x))
+    a = dja.dmap("abs", d)
+    got = a.localpart()
+    assert np.array_equal(np.signbit(got), np.signbit(np.abs(x)))
+    inv = dja.dmap("inv", a)
+    assert inv.localpart()[0] == np.inf   # not -inf
+    inv.close(); a.close(); d.close()
+    # reduce-side mapf too
+    s = dja.mapreduce("abs", "add", dja.distribute(np.array([-0.0, 1.0])))
+    assert s == 1.0
