@@ -161,3 +161,71 @@ def test_differential_fuzz(seed):
     for d in pool:
         d.close()
     assert dja.bytes_in_use() == 0
+
+
+@pytest.mark.parametrize("seed", [5, 6])
+def test_differential_fuzz_2d(seed):
+    """2-D structural ops: transpose, gather_box, diagonal scaling and
+    dims-reductions are data-movement/plumbing paths — fuzzed bit-exact
+    (mul is correctly rounded; transpose/gather move bits)."""
+    import ctypes
+    import distributedarrays_jl_amd as dja
+    from distributedarrays_jl_amd._ffi import lib, check
+    dja.comm.init()
+    nops = int(os.environ.get("FUZZ_OPS", "120"))
+    rng = np.random.default_rng(seed)
+    pool, mirror = [], []
+
+    def mk():
+        m = int(rng.integers(1, 200))
+        n = int(rng.integers(1, 200))
+        s = int(rng.integers(0, 2 ** 31))
+        d = dja.DArray((m, n), "f64")
+        d.rand_(seed_base=s)
+        pool.append(d)
+        mirror.append(np.asfortranarray(
+            philox.fill_uniform_f64(m * n, s).reshape(m, n, order="F")))
+
+    for _ in range(3):
+        mk()
+    for step in range(nops):
+        act = rng.integers(0, 5)
+        i = int(rng.integers(0, len(pool)))
+        x = mirror[i]
+        if act == 0 and len(pool) < 8:
+            mk()
+        elif act == 1:  # transpose (bit-exact data movement)
+            tx = dja.dtranspose(pool[i])
+            assert np.array_equal(tx.localpart(),
+                                  np.asfortranarray(x.T)), ("T", i)
+            tx.close()
+        elif act == 2:  # random sub-box gather (copy2d indexing)
+            m, n = x.shape
+            r0 = int(rng.integers(0, m)); r1 = int(rng.integers(r0, m)) + 1
+            c0 = int(rng.integers(0, n)); c1 = int(rng.integers(c0, n)) + 1
+            buf, shape = dja.gather_box(pool[i], [((r0, r1), (c0, c1))])
+            out = np.empty(shape, dtype=np.float64, order="F")
+            check(lib.da_d2h(buf.p, out.ctypes.data_as(ctypes.c_void_p),
+                             out.size * 8))
+            buf.free()
+            assert np.array_equal(out, x[r0:r1, c0:c1]), ("box", i)
+        elif act == 3:  # diagonal scaling in place (one rounded mul)
+            if rng.integers(0, 2):
+                dv = philox.fill_uniform_f64(x.shape[0], step + 1)
+                dja.ddiag_lmul(dv, pool[i])
+                mirror[i] = np.asfortranarray(dv[:, None] * x)
+            else:
+                dv = philox.fill_uniform_f64(x.shape[1], step + 1)
+                dja.ddiag_rmul(pool[i], dv)
+                mirror[i] = np.asfortranarray(x * dv[None, :])
+            assert np.array_equal(pool[i].localpart(), mirror[i]), ("dg", i)
+        elif act == 4:  # dims-reduction (tree order: tolerance)
+            axes = [(0,), (1,), (0, 1)][int(rng.integers(0, 3))]
+            R = dja.dsum_dims(pool[i], axes)
+            ref = x.sum(axis=axes, keepdims=True)
+            assert np.allclose(R.collect(), ref, rtol=1e-12,
+                               atol=1e-12), ("dims", i, axes)
+            R.close()
+    for d in pool:
+        d.close()
+    assert dja.bytes_in_use() == 0
