@@ -83,6 +83,8 @@ int da_init(int device, int rank, int nranks, const char* rccl_uid_path) {
                                          hipEventDisableTiming));
     DA_CHECK_HIP(hipEventCreateWithFlags(&st().ev_comm,
                                          hipEventDisableTiming));
+    DA_CHECK_HIP(hipMalloc(&st().red_ticket, sizeof(unsigned int)));
+    DA_CHECK_HIP(hipMemset(st().red_ticket, 0, sizeof(unsigned int)));
     st().device = device;
     st().rank = rank;
     st().nranks = nranks;
@@ -133,6 +135,8 @@ int da_shutdown(void) {
         st().bytes_in_use = 0;
         (void)pool_trim_locked();
     }
+    if (st().red_ticket) { (void)hipFree(st().red_ticket);
+                           st().red_ticket = nullptr; }
     if (st().ev_main) { (void)hipEventDestroy(st().ev_main); st().ev_main = nullptr; }
     if (st().ev_comm) { (void)hipEventDestroy(st().ev_comm); st().ev_comm = nullptr; }
     if (st().comm_stream) { (void)hipStreamDestroy(st().comm_stream);

@@ -34,6 +34,9 @@ struct State {
     // reduction partials buffer
     void* partials = nullptr;
     size_t partials_bytes = 0;
+    // fused-reduce fan-in ticket (dedicated 4 bytes, zeroed at init;
+    // must NOT live in `partials`, which dims-reduce reuses as slabs)
+    unsigned int* red_ticket = nullptr;
     std::mutex mem_mtx;
     std::unordered_map<void*, uint64_t> allocs;
     uint64_t bytes_in_use = 0;

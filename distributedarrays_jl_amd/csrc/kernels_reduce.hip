@@ -171,6 +171,63 @@ __global__ void reduce_stage1_v4(int mapop, int redop,
     if (threadIdx.x == 0) partials[blockIdx.x] = acc;
 }
 
+// Fused single-kernel reduce: stage-1 blocks publish their partial with
+// an agent-scope release and take a ticket; the LAST arriver acquires,
+// folds all partials and writes the result — the fan-in form of the
+// guide's split-K seam recipe (cdna_hip_programming.md §5, "In-launch
+// split-K reduction"; no grid-wide wait, so residency is irrelevant).
+// The ticket counter lives in the partials buffer, is zeroed at
+// allocation, and is reset by the last arriver; same-stream ordering
+// makes the reset visible to the next launch.
+template <typename T>
+__global__ void reduce_fused(int mapop, int redop, const T* __restrict__ src,
+                             uint64_t n, T* __restrict__ partials,
+                             unsigned int* __restrict__ ticket,
+                             T* __restrict__ out) {
+    __shared__ int is_last;
+    uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    T acc = RedIdent<T>::get(redop);
+    using V = T __attribute__((ext_vector_type(2)));
+    uint64_t nv = n / 2;
+    const V* sv = reinterpret_cast<const V*>(src);
+    for (uint64_t j = i; j < nv; j += stride) {
+        V v = sv[j];
+        acc = red_comb(redop, acc, mapf<T>(mapop, (T)v.x));
+        acc = red_comb(redop, acc, mapf<T>(mapop, (T)v.y));
+    }
+    for (uint64_t j = 2 * nv + i; j < n; j += stride)
+        acc = red_comb(redop, acc, mapf<T>(mapop, src[j]));
+    acc = block_reduce(redop, acc);
+    if (threadIdx.x == 0) partials[blockIdx.x] = acc;
+    // publish: wait the partial store, release, ticket (guide order:
+    // fence THEN fetch_add, with the asm vmcnt wait restated after the
+    // fence — ROCm 7.2 drops the post-wbl2 wait otherwise)
+    __asm__ volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __syncthreads();
+    if (threadIdx.x == 0) {
+        __builtin_amdgcn_fence(__ATOMIC_RELEASE, "agent");
+        __asm__ volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        unsigned int t = __hip_atomic_fetch_add(
+            ticket, 1u, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+        is_last = (t == gridDim.x - 1);
+    }
+    __syncthreads();
+    if (!is_last) return;
+    if (threadIdx.x == 0)
+        __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
+    __syncthreads();
+    T facc = RedIdent<T>::get(redop);
+    for (unsigned int j = threadIdx.x; j < gridDim.x; j += blockDim.x)
+        facc = red_comb(redop, facc, partials[j]);
+    __syncthreads();   // block_reduce LDS reused after the first pass
+    facc = block_reduce(redop, facc);
+    if (threadIdx.x == 0) {
+        out[0] = facc;
+        *ticket = 0;   // next launch on this stream sees 0
+    }
+}
+
 template <typename T>
 __global__ void reduce_stage2(int redop, const T* __restrict__ partials,
                               int np, T* __restrict__ out) {
@@ -191,20 +248,32 @@ static int do_reduce(int mapop, int redop, const T* src, uint64_t n,
     bool v4 = reduce_v4();
     uint64_t want = (n / (v4 ? 4 : 2) + RTPB - 1) / RTPB;
     int g = reduce_grid(want);
-    int rc = ensure_partials((RMAXB + 1) * sizeof(T));
+    int rc = ensure_partials((RMAXB + 1) * sizeof(double));
     if (rc) return rc;
     T* parts = (T*)st().partials;
-    T* dout = parts + RMAXB;
-    if (v4)
-        hipLaunchKernelGGL(reduce_stage1_v4<T>, dim3(g), dim3(RTPB), 0, s,
-                           mapop, redop, src, n, parts);
-    else
-        hipLaunchKernelGGL(reduce_stage1<T>, dim3(g), dim3(RTPB), 0, s,
-                           mapop, redop, src, n, parts);
-    DA_CHECK_HIP(hipGetLastError());
-    hipLaunchKernelGGL(reduce_stage2<T>, dim3(1), dim3(RTPB), 0, s,
-                       redop, parts, g, dout);
-    DA_CHECK_HIP(hipGetLastError());
+    T* dout = (T*)((double*)st().partials + RMAXB);
+    unsigned int* ticket = st().red_ticket;
+    static int fused = -1;
+    if (fused < 0) {
+        const char* e = getenv("DA_RED_FUSED");
+        fused = e ? atoi(e) : 1;
+    }
+    if (fused && !v4) {
+        hipLaunchKernelGGL(reduce_fused<T>, dim3(g), dim3(RTPB), 0, s,
+                           mapop, redop, src, n, parts, ticket, dout);
+        DA_CHECK_HIP(hipGetLastError());
+    } else {
+        if (v4)
+            hipLaunchKernelGGL(reduce_stage1_v4<T>, dim3(g), dim3(RTPB), 0,
+                               s, mapop, redop, src, n, parts);
+        else
+            hipLaunchKernelGGL(reduce_stage1<T>, dim3(g), dim3(RTPB), 0, s,
+                               mapop, redop, src, n, parts);
+        DA_CHECK_HIP(hipGetLastError());
+        hipLaunchKernelGGL(reduce_stage2<T>, dim3(1), dim3(RTPB), 0, s,
+                           redop, parts, g, dout);
+        DA_CHECK_HIP(hipGetLastError());
+    }
     DA_CHECK_HIP(hipMemcpyAsync(out_host, dout, sizeof(T),
                                 hipMemcpyDeviceToHost, s));
     DA_CHECK_HIP(hipStreamSynchronize(s));
