@@ -256,7 +256,7 @@ static int do_reduce(int mapop, int redop, const T* src, uint64_t n,
     static int fused = -1;
     if (fused < 0) {
         const char* e = getenv("DA_RED_FUSED");
-        fused = e ? atoi(e) : 1;
+        fused = e ? atoi(e) : 0;   // measured SLOWER at 2048 blocks (profiles)
     }
     if (fused && !v4) {
         hipLaunchKernelGGL(reduce_fused<T>, dim3(g), dim3(RTPB), 0, s,
