@@ -227,8 +227,8 @@ def main():
         cpu_baseline = {
             "value": n * 8.0 / tc / 1e9, "unit": "GB/s",
             "cores": int(clib.cb_num_threads()), "kind": "port",
-            "sample": "full 2^28-elem f64 sum, median of 5 (~%.2fs each)"
-                      % tc,
+            "sample": "full 2^28-elem f64 sum, median of 5 (%.0f ms each)"
+                      % (tc * 1e3),
         }
         del h
         log("[bench] cpu baseline: %.1f GB/s on %d cores"
