@@ -199,3 +199,128 @@ def test_gloo_matmul_world4(tmp_path):
 def test_gloo_matmul_world8(tmp_path):
     # 2x4 grid — the exact cfg-4 bench geometry (16384^2 over 8 GPUs)
     _spawn(_matmul_worker, tmp_path, world=8, extra=(8,))
+
+
+def _dims_reduce_worker(rank, tmpfile, q, world=4):
+    """world-4 numpy execution of ops.dreduce_dims' exchange schedule
+    (partials to the reduced-coord-0 owner, ascending-source combine)."""
+    try:
+        _init(rank, tmpfile, world)
+        dims, dist = (24, 20), (2, 2)
+        x = np.asfortranarray(philox.fill_uniform_f64(480, 9)
+                              .reshape(dims, order="F"))
+        idxs, cuts = pg.chunk_indices(dims, dist)
+        for red in [(0,), (1,), (0, 1)]:
+            loc = x[tuple(slice(lo, hi) for lo, hi in idxs[rank])]
+            part = loc.sum(axis=red, keepdims=True)
+            # owner groups exactly as ops.dreduce_dims builds them
+            groups = {}
+            for src in range(4):
+                sub = list(pg.grid_pos(src, dist))
+                for a in red:
+                    sub[a] = 0
+                groups.setdefault(pg.grid_rank(sub, dist), []).append(src)
+            sub = list(pg.grid_pos(rank, dist))
+            for a in red:
+                sub[a] = 0
+            my_owner = pg.grid_rank(sub, dist)
+            reqs, got = [], {}
+            if my_owner != rank:
+                t = torch.from_numpy(np.ascontiguousarray(part))
+                reqs.append((td.isend(t, my_owner, tag=hash(red) % 97),
+                             None, t))
+            if rank in groups:
+                for src in groups[rank]:
+                    if src != rank:
+                        t = torch.zeros(part.shape, dtype=torch.float64)
+                        reqs.append((td.irecv(t, src, tag=hash(red) % 97),
+                                     src, t))
+            for rq, key, t in reqs:
+                rq.wait()
+                if key is not None:
+                    got[key] = t.numpy()
+            if rank in groups:
+                acc = None
+                for src in groups[rank]:      # ascending source order
+                    p = part if src == rank else got[src]
+                    acc = p.copy() if acc is None else acc + p
+                # owner's slab must equal the reference reduction
+                my_nonred = tuple(
+                    slice(0, 1) if a in red else
+                    slice(idxs[rank][a][0], idxs[rank][a][1])
+                    for a in range(2))
+                ref = x.sum(axis=red, keepdims=True)[my_nonred]
+                assert np.allclose(acc, ref, rtol=1e-12), red
+        q.put((rank, True, None))
+    except Exception as e:  # pragma: no cover
+        q.put((rank, False, repr(e)))
+    finally:
+        if td.is_initialized():
+            td.destroy_process_group()
+
+
+def _sort_worker(rank, tmpfile, q, world=4):
+    """world-4 numpy execution of ops.dsort's segment all-to-all."""
+    try:
+        _init(rank, tmpfile, world)
+        n = 4003
+        idxs, _ = pg.chunk_indices((n,), (world,))
+        lo, hi = idxs[rank][0]
+        full = philox.fill_uniform_f64(n, 33)
+        mine = np.sort(full[lo:hi])
+        # sampled splitters (same protocol as ops.dsort)
+        s = min(64, mine.size)
+        sel = ((np.arange(s) + 0.5) * mine.size / s).astype(np.int64)
+        samples = mine[sel]
+        gathered = [None] * world
+        td.all_gather_object(gathered, samples)
+        allsamp = np.sort(np.concatenate(gathered))
+        splitters = allsamp[[(i + 1) * allsamp.size // world
+                             for i in range(world - 1)]]
+        edges = [0] + [int(np.searchsorted(mine, sp)) for sp in
+                       splitters] + [mine.size]
+        segs = [edges[j + 1] - edges[j] for j in range(world)]
+        allsegs = [None] * world
+        td.all_gather_object(allsegs, segs)
+        recv_sizes = [allsegs[src][rank] for src in range(world)]
+        reqs, got = [], {}
+        for dst in range(world):
+            if dst != rank and segs[dst]:
+                t = torch.from_numpy(
+                    np.ascontiguousarray(mine[edges[dst]:edges[dst + 1]]))
+                reqs.append((td.isend(t, dst, tag=dst), None, t))
+        for src in range(world):
+            if src != rank and recv_sizes[src]:
+                t = torch.zeros(recv_sizes[src], dtype=torch.float64)
+                reqs.append((td.irecv(t, src, tag=rank), src, t))
+        for rq, key, t in reqs:
+            rq.wait()
+            if key is not None:
+                got[key] = t.numpy()
+        parts = []
+        for src in range(world):
+            if src == rank:
+                parts.append(mine[edges[rank]:edges[rank + 1]])
+            elif src in got:
+                parts.append(got[src])
+        result = np.sort(np.concatenate(parts)) if parts else \
+            np.empty(0)
+        gathered = [None] * world
+        td.all_gather_object(gathered, result)
+        assert np.array_equal(np.concatenate(gathered), np.sort(full))
+        q.put((rank, True, None))
+    except Exception as e:  # pragma: no cover
+        q.put((rank, False, repr(e)))
+    finally:
+        if td.is_initialized():
+            td.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_gloo_dims_reduce_world4(tmp_path):
+    _spawn(_dims_reduce_worker, tmp_path, world=4, extra=(4,))
+
+
+@pytest.mark.timeout(300)
+def test_gloo_sort_world4(tmp_path):
+    _spawn(_sort_worker, tmp_path, world=4, extra=(4,))
