@@ -242,7 +242,10 @@ class DArray:
         the reference's remotecall gather (darray.jl:574ff)."""
         npdt = np.dtype(NUMPY_DTYPES[self.dtype])
         local = self.localpart()
-        if self.nchunks == 1 and self.lchunk is not None:
+        if self.nranks == 1:
+            # (the fast path must be gated on the WORLD size, not the
+            # chunk count: at nranks>1 every rank must join the gather
+            # below or the collective mismatches)
             out = np.zeros(self.dims, dtype=npdt, order="F")
             if self.lnumel:
                 sl = tuple(slice(lo, hi) for lo, hi in self.lidx)
