@@ -50,3 +50,39 @@ def test_philox_offset_consistency(seed, n, off):
                                                 offset=off + n // 2)])
     assert np.array_equal(a, b)
     assert a.min() >= 0.0 and a.max() < 1.0
+
+
+@given(st.integers(0, 3000), st.integers(1, 8), st.integers(0, 2**31 - 1))
+@settings(max_examples=60, deadline=None)
+def test_samplesort_sim_property(n, P, seed):
+    from tests.test_sort import simulate_samplesort
+    x = philox.fill_uniform_f64(n, seed) if n else np.empty(0)
+    got, sizes = simulate_samplesort(x, P)
+    assert np.array_equal(got, np.sort(x))
+    assert sum(sizes) == n and len(sizes) == P
+
+
+@given(st.integers(1, 60), st.integers(1, 60),
+       st.integers(1, 8), st.integers(0, 2**31 - 1))
+@settings(max_examples=60, deadline=None)
+def test_halo_plan_property(m, n, nr, seed):
+    from distributedarrays_jl_amd import plan
+    rng = np.random.default_rng(seed)
+    dist = pg.defaultdist((m, n), nr)
+    idxs, _ = pg.chunk_indices((m, n), dist)
+    nchunks = len(idxs)
+    ranks = list(range(nchunks))
+    r0 = int(rng.integers(0, m)); r1 = int(rng.integers(r0, m)) + 1
+    c0 = int(rng.integers(0, n)); c1 = int(rng.integers(c0, n)) + 1
+    boxes = [None] * nchunks
+    boxes[int(rng.integers(0, nchunks))] = ((r0, r1), (c0, c1))
+    pieces = plan.halo_plan(idxs, ranks, boxes)
+    cover = np.zeros((m, n), dtype=int)
+    for (src, dst, b) in pieces:
+        sl = tuple(slice(lo, hi) for lo, hi in b)
+        cover[sl] += 1
+        # every piece must lie inside its source chunk
+        for (lo, hi), (clo, chi) in zip(b, idxs[src]):
+            assert clo <= lo and hi <= chi
+    assert (cover[r0:r1, c0:c1] == 1).all()
+    assert cover.sum() == (r1 - r0) * (c1 - c0)
