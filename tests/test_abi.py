@@ -72,3 +72,13 @@ def test_errstr_without_init(built):
     assert rc < 0
     built.da_errstr.restype = ctypes.c_char_p
     assert b"da_init" in built.da_errstr(rc)
+
+
+def test_integration_stubs_exist(built):
+    """Every ccall symbol named in INTEGRATION.md's Julia stub block must
+    be exported by the library."""
+    txt = open(os.path.join(ROOT, "INTEGRATION.md")).read()
+    syms = set(re.findall(r":(da_\w+)", txt))
+    assert len(syms) >= 15
+    missing = [s for s in syms if not hasattr(built, s)]
+    assert not missing, missing
