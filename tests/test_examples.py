@@ -14,3 +14,12 @@ def test_power_iteration():
     from examples.power_iteration import dominant_eig
     lam = dominant_eig(512, iters=25)
     assert abs(lam - 256) < 10   # uniform matrix: lambda ~ n/2
+
+
+def test_column_stats():
+    from examples.column_stats import column_stats
+    means, mx, finite, med = column_stats(2048, 256)
+    assert finite
+    assert abs(float(means.mean()) - 0.5) < 0.01
+    assert float(mx.min()) > 0.99
+    assert abs(med - 0.5) < 0.05
