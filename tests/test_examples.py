@@ -23,3 +23,10 @@ def test_column_stats():
     assert abs(float(means.mean()) - 0.5) < 0.01
     assert float(mx.min()) > 0.99
     assert abs(med - 0.5) < 0.05
+
+
+def test_game_of_life():
+    from examples.game_of_life import run
+    ok, pop = run(128, 8)
+    assert ok, "life diverged from numpy reference"
+    assert pop > 0
