@@ -651,3 +651,12 @@ def test_abs_signed_zero(dja):
     # reduce-side mapf too
     s = dja.mapreduce("abs", "add", dja.distribute(np.array([-0.0, 1.0])))
     assert s == 1.0
+
+
+def test_device_props(dja):
+    from distributedarrays_jl_amd._ffi import lib, check
+    name = ctypes.create_string_buffer(128)
+    hbm = ctypes.c_uint64()
+    check(lib.da_device_props(name, 128, ctypes.byref(hbm)))
+    assert hbm.value > 200 * 1024 ** 3   # 288 GB HBM3E
+    assert len(name.value) > 0
