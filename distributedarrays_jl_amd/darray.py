@@ -167,8 +167,12 @@ class DArray:
         return len(self.dims)
 
     def samedist(self, other):
-        return (self.dims == other.dims and self.dist == other.dist
-                and self.dtype == other.dtype)
+        """Identical chunk layout: same boxes AND owners (comparing idxs,
+        not dist — a ragged from_chunk_sizes DVector shares dist with an
+        evenly-cut one but not the cut points)."""
+        return (self.dims == other.dims and self.dtype == other.dtype
+                and list(self.idxs) == list(other.idxs)
+                and self.ranks == other.ranks)
 
     def _ptr(self):
         if self._chunk is None:
@@ -313,10 +317,30 @@ class DArray:
         return value
 
     def __eq__(self, other):
+        """== is elementwise-all equality (test/darray.jl:84-129);
+        mismatched cuts localize `other` onto self's boxes via the
+        makelocal gather (collective — the branch is metadata-determined
+        so every rank takes the same path)."""
         if isinstance(other, DArray):
             if self.dims != other.dims:
                 return False
-            a, b = self.localpart(), other_aligned_localpart(self, other)
+            a = self.localpart()
+            if self.samedist(other):
+                b = other.localpart()
+            else:
+                from . import ops
+                buf, shape = ops.gather_box(other, ops._dest_boxes(self))
+                if buf is not None:
+                    b = np.empty(shape, order="F",
+                                 dtype=np.dtype(NUMPY_DTYPES[other.dtype]))
+                    if b.size:
+                        check(lib.da_d2h(
+                            buf.p, b.ctypes.data_as(ctypes.c_void_p),
+                            b.size * DTYPE_SIZE[other.dtype]))
+                    buf.free()
+                else:
+                    b = np.empty(self.lshape, order="F",
+                                 dtype=np.dtype(NUMPY_DTYPES[other.dtype]))
             import torch.distributed as td
             same = bool(np.array_equal(a, b))
             if self.nranks > 1 and td.is_initialized():
@@ -333,12 +357,6 @@ class DArray:
         return ("DArray(dims=%r, dist=%r, dtype=%s, rank=%d/%d, lshape=%r)"
                 % (self.dims, self.dist, self.dtype, self.rank, self.nranks,
                    self.lshape))
-
-
-def other_aligned_localpart(d, other):
-    if d.dist != other.dist or d.dims != other.dims:
-        raise _ffi.DArrayError("== needs aligned distributions (round 1)")
-    return other.localpart()
 
 
 # ---- convenience constructors (darray.jl:460-532) ----

@@ -16,24 +16,49 @@ from ._opcodes import (DTYPES, DTYPE_SIZE, NUMPY_DTYPES, MAP_OP, MAP2_OP,
 from .darray import DArray
 
 
+def _same_layout(d0, d):
+    """True when the two DArrays have identical chunk layout: same global
+    dims/dtype, the same cut boxes AND the same chunk->rank owners.
+    Comparing d.idxs (not d.dist) matters: a ragged DVector from
+    DArray.from_chunk_sizes has dist=(nr,) like an evenly-cut one, but
+    per-rank lnumel differs — treating those as aligned would read past
+    the smaller chunk."""
+    return (d.dims == d0.dims and d.dtype == d0.dtype
+            and list(d.idxs) == list(d0.idxs) and d.ranks == d0.ranks)
+
+
 def _aligned(*ds):
     d0 = ds[0]
     for d in ds[1:]:
-        if (d.dims != d0.dims or d.dist != d0.dist
-                or d.dtype != d0.dtype or d.ranks != d0.ranks):
+        if not _same_layout(d0, d):
             raise DArrayError(
-                "operands must share dims/dist/dtype/owners (aligned "
-                "cuts); use map_general/broadcast_fma_general for "
-                "mismatched cuts")
+                "operands must share dims/dtype/cuts/owners (aligned "
+                "chunks); mismatched-cuts operands route through the "
+                "makelocal gather (map_general/map2_general/"
+                "broadcast_fma_general)")
     return d0
+
+
+def _check_i64_scalar(dtype, c):
+    """The ABI carries scalars as double; |i64| > 2^53 would round
+    silently (the fill_ guard, generalized — ADVICE r1)."""
+    if dtype == "i64":
+        ci = int(c)
+        if abs(ci) > (1 << 53) or ci != c:
+            raise DArrayError(
+                "i64 scalar %r not exactly representable through the "
+                "double ABI parameter" % (c,))
 
 
 # ------------------------------------------------------- map / broadcast
 def map_(op, dest, src):
-    """map!(f, dest, src) — mapreduce.jl:5-12 (aligned cuts)."""
-    _aligned(dest, src)
+    """map!(f, dest, src) — mapreduce.jl:5-12.  Mismatched cuts route
+    transparently through the makelocal gather (mapreduce.jl:8), like
+    the reference."""
     if dest.dtype == "i64" and op not in I64_MAP_OPS:
         raise DArrayError("op %r invalid for i64" % op)
+    if not _same_layout(dest, src):
+        return map_general(op, dest, src)
     if dest.lnumel:
         check(lib.da_map(MAP_OP[op], dest._ptr(), src._ptr(),
                          dest.lnumel, DTYPES[dest.dtype]))
@@ -46,10 +71,12 @@ def dmap(op, src):
 
 
 def map2_(op, dest, a, b):
-    """dest .= f.(a, b) for the binary table (mapreduce.jl:180-189)."""
-    _aligned(dest, a, b)
+    """dest .= f.(a, b) for the binary table (mapreduce.jl:180-189).
+    Mismatched cuts localize per-operand (bclocal, broadcast.jl:140-152)."""
     if dest.dtype == "i64" and op not in I64_MAP2_OPS:
         raise DArrayError("op %r invalid for i64" % op)
+    if not (_same_layout(dest, a) and _same_layout(dest, b)):
+        return map2_general(op, dest, a, b)
     if dest.lnumel:
         check(lib.da_map2(MAP2_OP[op], dest._ptr(), a._ptr(), b._ptr(),
                           dest.lnumel, DTYPES[dest.dtype]))
@@ -64,9 +91,19 @@ def map2_scalar_(op, dest, src, c, reverse=False):
     """dest .= f.(src, c) (or f.(c, src)) — scalar broadcast argument
     (broadcast.jl:124-133: singletons are not distributed).  Covers the
     cfg-1 plumbing op D .+ 1."""
-    _aligned(dest, src)
     if dest.dtype == "i64" and op not in I64_MAP2_OPS:
         raise DArrayError("op %r invalid for i64" % op)
+    _check_i64_scalar(dest.dtype, c)
+    if not _same_layout(dest, src):
+        buf, _ = gather_box(src, _dest_boxes(dest))
+        if buf is not None and dest.lnumel:
+            check(lib.da_map2_scalar(MAP2_OP[op], dest._ptr(), buf.p,
+                                     float(c), 1 if reverse else 0,
+                                     dest.lnumel, DTYPES[dest.dtype]))
+            check(lib.da_synchronize())
+        if buf is not None:
+            buf.free()
+        return dest
     if dest.lnumel:
         check(lib.da_map2_scalar(MAP2_OP[op], dest._ptr(), src._ptr(),
                                  float(c), 1 if reverse else 0,
@@ -80,8 +117,10 @@ def elementwise_scalar(op, a, c, reverse=False):
 
 def broadcast_fma(dest, a, b, c):
     """D .= A .* B .+ c — the fused cfg-3 broadcast (broadcast.jl:65-85;
-    aligned same-cuts args need zero communication)."""
-    _aligned(dest, a, b)
+    aligned same-cuts args need zero communication).
+    Mismatched cuts route through the bclocal/makelocal gather."""
+    if not (_same_layout(dest, a) and _same_layout(dest, b)):
+        return broadcast_fma_general(dest, a, b, c)
     if dest.lnumel:
         check(lib.da_bcast_fma(dest._ptr(), a._ptr(), b._ptr(), float(c),
                                dest.lnumel, DTYPES[dest.dtype]))
@@ -98,8 +137,20 @@ def axpy_(alpha, x, y):
 
 
 def add_(dest, src, scale=1.0):
-    """add! — linalg.jl:62-76."""
-    _aligned(dest, src)
+    """add! — linalg.jl:62-76.  Mismatched cuts gather src first
+    (the fetch of linalg.jl:243-251 partials is the aligned case)."""
+    _check_i64_scalar(dest.dtype, scale)
+    if not _same_layout(dest, src):
+        if dest.dims != src.dims or dest.dtype != src.dtype:
+            raise DArrayError("add_: dims/dtype mismatch")
+        buf, _ = gather_box(src, _dest_boxes(dest))
+        if buf is not None and dest.lnumel:
+            check(lib.da_add(dest._ptr(), buf.p, float(scale),
+                             dest.lnumel, DTYPES[dest.dtype]))
+            check(lib.da_synchronize())
+        if buf is not None:
+            buf.free()
+        return dest
     if dest.lnumel:
         check(lib.da_add(dest._ptr(), src._ptr(), float(scale),
                          dest.lnumel, DTYPES[dest.dtype]))
@@ -108,6 +159,7 @@ def add_(dest, src, scale=1.0):
 
 def scale_(a, s):
     """rmul! — linalg.jl:54-59."""
+    _check_i64_scalar(a.dtype, s)
     if a.lnumel:
         check(lib.da_scale(a._ptr(), float(s), a.lnumel, DTYPES[a.dtype]))
     return a
@@ -187,8 +239,10 @@ def dany(pred, d):
 
 
 def ddot(x, y):
-    """dot — linalg.jl:36-45 (aligned cuts: local fused path)."""
-    _aligned(x, y)
+    """dot — linalg.jl:36-45: aligned cuts multiply locally; mismatched
+    cuts localize y via makelocal (linalg.jl:42) inside map2_."""
+    if x.dims != y.dims or x.dtype != y.dtype:
+        raise DArrayError("ddot: dims/dtype mismatch")
     tmp = elementwise("mul", x, y)
     try:
         return mapreduce("identity", "add", tmp)
@@ -253,6 +307,14 @@ def dmatmul(A, B, alpha=1.0):
     if A.ndims != 2 or B.ndims != 2 or A.dims[1] != B.dims[0]:
         raise DArrayError("dmatmul: shape mismatch %r x %r"
                           % (A.dims, B.dims))
+    if (A.ranks != list(range(A.nchunks))
+            or B.ranks != list(range(B.nchunks))):
+        # the slab/partial plans treat chunk indices as rank ids;
+        # non-identity owners (e.g. a dims-reduction result) would
+        # silently drop the off-grid rank's data
+        raise DArrayError(
+            "dmatmul: operands must have identity chunk->rank mapping "
+            "(copy() a reduction result first)")
     if A.dtype == "f64":
         gemm_fn = lib.da_gemm_f64
     elif A.dtype == "f32":
@@ -276,61 +338,63 @@ def dmatmul(A, B, alpha=1.0):
     my_recvs = [p for p in pieces if p[1] == r and p[0] != r]
     my_local = [p for p in pieces if p[0] == r and p[1] == r]
 
+    # B-slab exchange: ANY rank owning B pieces packs+sends (a B owner
+    # outside A's process grid — the arbitrary-AbstractMatrix-B case of
+    # linalg.jl:211-226 — has pos None and only sends); A-grid ranks
+    # receive/unpack into their row slab.
     slab = None
+    srows = rlo = 0
     sendbufs, recvbufs = [], []
     if pos is not None:
         i, j = pos
         rlo, rhi = plan.slab_rows(A.cuts[1], j)
         srows = rhi - rlo
-        slab = _Buf(srows * n * esz)
-
-        # pack my outgoing pieces from my B block (da_copy2d: column-major
-        # sub-block -> contiguous)
-        if my_sends or my_recvs or my_local:
-            Brows = B.lshape[0] if B.lnumel else 0
-            for (src, dst, rows, cols) in my_sends:
-                prows = rows[1] - rows[0]
-                pcols = cols[1] - cols[0]
-                buf = _Buf(prows * pcols * esz)
-                off = ((rows[0] - B.lidx[0][0])
-                       + (cols[0] - B.lidx[1][0]) * Brows) * esz
-                _copy2d(buf.p, prows * esz,
-                        ctypes.c_void_p(B._ptr().value + off), Brows * esz,
-                        prows * esz, pcols)
-                sendbufs.append(((src, dst, rows, cols), buf))
-            for (src, dst, rows, cols) in my_recvs:
-                prows = rows[1] - rows[0]
-                pcols = cols[1] - cols[0]
-                recvbufs.append(((src, dst, rows, cols),
-                                 _Buf(prows * pcols * esz)))
-            # exchange (grouped so send/recv pairs match globally)
-            if my_sends or my_recvs:
-                check(lib.da_group_start())
-                for (src, dst, rows, cols), buf in sendbufs:
-                    nb = (rows[1] - rows[0]) * (cols[1] - cols[0]) * esz
-                    check(lib.da_send(buf.p, nb, dst))
-                for (src, dst, rows, cols), buf in recvbufs:
-                    nb = (rows[1] - rows[0]) * (cols[1] - cols[0]) * esz
-                    check(lib.da_recv(buf.p, nb, src))
-                check(lib.da_group_end())
-            # unpack into the slab (stream-ordered after the group)
-            for (src, dst, rows, cols) in my_local:
-                prows = rows[1] - rows[0]
-                pcols = cols[1] - cols[0]
-                off = ((rows[0] - B.lidx[0][0])
-                       + (cols[0] - B.lidx[1][0]) * (B.lshape[0])) * esz
-                doff = ((rows[0] - rlo) + cols[0] * srows) * esz
-                _copy2d(slab.at(doff), srows * esz,
-                        ctypes.c_void_p(B._ptr().value + off),
-                        B.lshape[0] * esz, prows * esz, pcols)
-            for (src, dst, rows, cols), buf in recvbufs:
-                prows = rows[1] - rows[0]
-                pcols = cols[1] - cols[0]
-                doff = ((rows[0] - rlo) + cols[0] * srows) * esz
-                _copy2d(slab.at(doff), srows * esz, buf.p, prows * esz,
-                        prows * esz, pcols)
-    elif my_sends or my_recvs or my_local:
-        raise DArrayError("B owner outside A's grid unsupported (round 1)")
+        slab = _Buf(max(srows * n, 1) * esz)
+    # pack my outgoing pieces from my B block (da_copy2d: column-major
+    # sub-block -> contiguous)
+    Brows = B.lshape[0] if B.lnumel else 0
+    for (src, dst, rows, cols) in my_sends:
+        prows = rows[1] - rows[0]
+        pcols = cols[1] - cols[0]
+        buf = _Buf(prows * pcols * esz)
+        off = ((rows[0] - B.lidx[0][0])
+               + (cols[0] - B.lidx[1][0]) * Brows) * esz
+        _copy2d(buf.p, prows * esz,
+                ctypes.c_void_p(B._ptr().value + off), Brows * esz,
+                prows * esz, pcols)
+        sendbufs.append(((src, dst, rows, cols), buf))
+    for (src, dst, rows, cols) in my_recvs:
+        prows = rows[1] - rows[0]
+        pcols = cols[1] - cols[0]
+        recvbufs.append(((src, dst, rows, cols),
+                         _Buf(prows * pcols * esz)))
+    # exchange (grouped so send/recv pairs match globally)
+    if my_sends or my_recvs:
+        check(lib.da_group_start())
+        for (src, dst, rows, cols), buf in sendbufs:
+            nb = (rows[1] - rows[0]) * (cols[1] - cols[0]) * esz
+            check(lib.da_send(buf.p, nb, dst))
+        for (src, dst, rows, cols), buf in recvbufs:
+            nb = (rows[1] - rows[0]) * (cols[1] - cols[0]) * esz
+            check(lib.da_recv(buf.p, nb, src))
+        check(lib.da_group_end())
+    if pos is not None:
+        # unpack into the slab (stream-ordered after the group)
+        for (src, dst, rows, cols) in my_local:
+            prows = rows[1] - rows[0]
+            pcols = cols[1] - cols[0]
+            off = ((rows[0] - B.lidx[0][0])
+                   + (cols[0] - B.lidx[1][0]) * (B.lshape[0])) * esz
+            doff = ((rows[0] - rlo) + cols[0] * srows) * esz
+            _copy2d(slab.at(doff), srows * esz,
+                    ctypes.c_void_p(B._ptr().value + off),
+                    B.lshape[0] * esz, prows * esz, pcols)
+        for (src, dst, rows, cols), buf in recvbufs:
+            prows = rows[1] - rows[0]
+            pcols = cols[1] - cols[0]
+            doff = ((rows[0] - rlo) + cols[0] * srows) * esz
+            _copy2d(slab.at(doff), srows * esz, buf.p, prows * esz,
+                    prows * esz, pcols)
 
     # local partial GEMMs fused with the per-k partial exchange
     # (linalg.jl:218-251).  With DA_MM_OVERLAP (default on) each k's
@@ -352,19 +416,29 @@ def dmatmul(A, B, alpha=1.0):
     if overlap:
         check(lib.da_p2p_stream(1))
     for k in range(K):
-        if pos is not None and A.lnumel:
+        if pos is not None:
+            # a partial per k even when the local A chunk is empty
+            # (zero-size k-cut): its send/recv peers index partials[k],
+            # and an empty k-sum contributes zeros
             mloc = A.lshape[0]
             kloc = A.lshape[1]
             clo, chi = ccols[k]
             nk = chi - clo
-            pk = _Buf(mloc * nk * esz)
-            check(gemm_fn(pk.p, A._ptr(), slab.at(clo * kloc * esz),
-                          mloc, nk, kloc, mloc, kloc, mloc, 1.0, 0.0))
+            pk = _Buf(max(mloc * nk, 1) * esz)
+            if A.lnumel:
+                check(gemm_fn(pk.p, A._ptr(), slab.at(clo * kloc * esz),
+                              mloc, nk, kloc, mloc, kloc, mloc, 1.0, 0.0))
+            elif mloc * nk:
+                check(lib.da_fill(pk.p, 0.0, mloc * nk, DTYPES[A.dtype]))
             partials.append(pk)
         if overlap:
             check(lib.da_comm_after_compute())
-            sends_k = [mv for mv in my_psends if mv[2] == k]
-            recvs_k = [mv for mv in my_precvs if mv[2] == k]
+            # zero-size partials are skipped consistently on both sides:
+            # sender nb == A.lshape[0]*width == C-owner's C.lnumel
+            sends_k = [mv for mv in my_psends if mv[2] == k
+                       and A.lshape[0] * (ccols[k][1] - ccols[k][0])]
+            recvs_k = [mv for mv in my_precvs if mv[2] == k
+                       and C.lnumel]
             if sends_k or recvs_k:
                 check(lib.da_group_start())
                 for (src, dst, kk) in sends_k:
@@ -377,16 +451,20 @@ def dmatmul(A, B, alpha=1.0):
     if overlap:
         check(lib.da_main_after_comm())
         check(lib.da_p2p_stream(0))
-    elif my_psends or my_precvs:
-        check(lib.da_group_start())
-        for (src, dst, k) in my_psends:
-            nb = A.lshape[0] * (ccols[k][1] - ccols[k][0]) * esz
-            check(lib.da_send(partials[k].p, nb, dst))
-        for (src, dst, k) in my_precvs:
-            buf = precv[(src, k)]
-            nb = C.lshape[0] * C.lshape[1] * esz
-            check(lib.da_recv(buf.p, nb, src))
-        check(lib.da_group_end())
+    else:
+        psends = [mv for mv in my_psends
+                  if A.lshape[0] * (ccols[mv[2]][1] - ccols[mv[2]][0])]
+        precvs = [mv for mv in my_precvs if C.lnumel]
+        if psends or precvs:
+            check(lib.da_group_start())
+            for (src, dst, k) in psends:
+                nb = A.lshape[0] * (ccols[k][1] - ccols[k][0]) * esz
+                check(lib.da_send(partials[k].p, nb, dst))
+            for (src, dst, k) in precvs:
+                buf = precv[(src, k)]
+                nb = C.lshape[0] * C.lshape[1] * esz
+                check(lib.da_recv(buf.p, nb, src))
+            check(lib.da_group_end())
     if r < I * K and C.lnumel:
         i, myk = r % I, r // I
         for j in plan.accumulate_order(J):
@@ -561,6 +639,7 @@ def dmean_dims(d, dims):
     """mean(D; dims) rides the sum path (ext/StatisticsExt.jl:6)."""
     if isinstance(dims, int):
         dims = (dims,)
+    dims = tuple(sorted(set(int(a) for a in dims)))  # match dreduce_dims
     R = dsum_dims(d, dims)
     nred = 1
     for a in dims:
@@ -578,6 +657,10 @@ def dmatvec(A, x, alpha=1.0):
     import numpy as np
     if A.dtype != "f64" or A.ndims != 2:
         raise DArrayError("dmatvec: 2-D f64 only")
+    if A.ranks != list(range(A.nchunks)):
+        raise DArrayError(
+            "dmatvec: A must have identity chunk->rank mapping "
+            "(copy() a reduction result first)")
     x = np.ascontiguousarray(np.asarray(x, dtype=np.float64))
     if x.shape != (A.dims[1],):
         raise DArrayError("dmatvec: x length %d != %d"
@@ -773,7 +856,7 @@ def map_general(op, dest, src):
     makelocal): gather src's piece of dest's index box, then map."""
     if dest.dims != src.dims or dest.dtype != src.dtype:
         raise DArrayError("map_general: dims/dtype mismatch")
-    if dest.dist == src.dist and dest.ranks == src.ranks:
+    if _same_layout(dest, src):
         return map_(op, dest, src)
     buf, shape = gather_box(src, _dest_boxes(dest))
     if buf is not None:
@@ -784,20 +867,48 @@ def map_general(op, dest, src):
     return dest
 
 
+def map2_general(op, dest, a, b):
+    """dest .= f.(a, b) with arbitrary (mismatched) cuts — per-operand
+    bclocal/makelocal localisation (broadcast.jl:140-152): only the
+    operands whose layout differs from dest's are gathered."""
+    if (dest.dims != a.dims or dest.dims != b.dims
+            or dest.dtype != a.dtype or dest.dtype != b.dtype):
+        raise DArrayError("map2_general: dims/dtype mismatch")
+    if _same_layout(dest, a) and _same_layout(dest, b):
+        return map2_(op, dest, a, b)
+    boxes = _dest_boxes(dest)
+    abuf = bbuf = None
+    if not _same_layout(dest, a):
+        abuf, _ = gather_box(a, boxes)
+    if not _same_layout(dest, b):
+        bbuf, _ = gather_box(b, boxes)
+    if dest.lnumel:
+        ap = abuf.p if abuf is not None else a._ptr()
+        bp = bbuf.p if bbuf is not None else b._ptr()
+        check(lib.da_map2(MAP2_OP[op], dest._ptr(), ap, bp,
+                          dest.lnumel, DTYPES[dest.dtype]))
+    check(lib.da_synchronize())
+    if abuf is not None:
+        abuf.free()
+    if bbuf is not None:
+        bbuf.free()
+    return dest
+
+
 def broadcast_fma_general(dest, a, b, c):
     """D .= A .* B .+ c with arbitrary (mismatched) cuts — the
     bclocal/makelocal localisation of broadcast.jl:65-85."""
-    if dest.samedist(a) and dest.samedist(b):
+    if _same_layout(dest, a) and _same_layout(dest, b):
         return broadcast_fma(dest, a, b, c)
     boxes = _dest_boxes(dest)
     abuf = bbuf = None
-    if not dest.samedist(a):
+    if not _same_layout(dest, a):
         abuf, _ = gather_box(a, boxes)
-    if not dest.samedist(b):
+    if not _same_layout(dest, b):
         bbuf, _ = gather_box(b, boxes)
-    ap = abuf.p if abuf is not None else a._ptr()
-    bp = bbuf.p if bbuf is not None else b._ptr()
     if dest.lnumel:
+        ap = abuf.p if abuf is not None else a._ptr()
+        bp = bbuf.p if bbuf is not None else b._ptr()
         check(lib.da_bcast_fma(dest._ptr(), ap, bp, float(c),
                                dest.lnumel, DTYPES[dest.dtype]))
     check(lib.da_synchronize())
