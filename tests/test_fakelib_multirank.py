@@ -1,0 +1,358 @@
+"""Multi-rank CPU execution of the REAL package code (ops.py/darray.py/
+spmd.py, unmodified) over the fakelib ABI: numpy chunks + gloo transport
+with RCCL grouped-p2p pairing semantics (see fakelib.py header).
+
+These complement tests/test_gloo.py: there, the *schedules* are
+re-executed by test code; here, the product's own orchestration code
+runs at world_size 2..4 — the closest this 1-GPU pool gets to the
+driver's 8-GPU round-end run.  Every scenario ends with d_closeall() +
+a zero-leak assertion over live ABI allocations, mirroring the
+reference's registry-empty check (test/darray.jl:1079-1086)."""
+import os
+
+import numpy as np
+import pytest
+import torch.multiprocessing as mp
+
+
+def _run_scenario(rank, tmpfile, q, world, name):
+    try:
+        os.environ["RANK"] = str(rank)
+        os.environ["WORLD_SIZE"] = str(world)
+        os.environ["MASTER_ADDR"] = "127.0.0.1"
+        import torch.distributed as td
+        td.init_process_group("gloo", init_method="file://%s" % tmpfile,
+                              rank=rank, world_size=world)
+        import fakelib
+        fake = fakelib.install()
+        import distributedarrays_jl_amd as dja
+        dja.comm.init()
+        globals()["_scenario_" + name](rank, world, dja)
+        dja.d_closeall()
+        assert fake.user_bytes == 0, \
+            "leaked %d bytes of ABI allocations" % fake.user_bytes
+        q.put((rank, True, None))
+    except Exception as e:
+        import traceback
+        q.put((rank, False, traceback.format_exc()[-2000:]))
+    finally:
+        import torch.distributed as td
+        if td.is_initialized():
+            td.destroy_process_group()
+
+
+def _spawn(tmp_path, world, name):
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    tmpfile = str(tmp_path / "rdv")
+    procs = [ctx.Process(target=_run_scenario,
+                         args=(r, tmpfile, q, world, name))
+             for r in range(world)]
+    for p in procs:
+        p.start()
+    results = [q.get(timeout=240) for _ in range(world)]
+    for p in procs:
+        p.join(timeout=60)
+    for rank, ok, err in sorted(results):
+        assert ok, "rank %d failed:\n%s" % (rank, err)
+
+
+# --------------------------------------------------------------- helpers
+def _global_f64(n, seed=7):
+    from oracle import philox
+    return philox.fill_uniform_f64(n, seed)
+
+
+def _slice_set(d, full):
+    """set_localpart from a global array (F-order slices)."""
+    sl = tuple(slice(lo, hi) for lo, hi in d.lidx)
+    d.set_localpart(np.asfortranarray(np.asarray(full)[sl]))
+    return d
+
+
+# -------------------------------------------------------------- scenarios
+def _scenario_basic(rank, world, dja):
+    from oracle import ops as oops, philox
+    n = 1000
+    D = dja.drand((n,), "f64")
+    # per-rank philox chunks, same protocol as the oracle fold
+    chunks = [philox.fill_uniform_f64(
+        d1 - d0, seed=1234 + r) for r, ((d0, d1),) in enumerate(D.idxs)]
+    ref = oops.oracle_reduce("identity", "add", chunks)
+    got = dja.dsum(D)
+    assert abs(got - ref) <= 1e-12 * abs(ref)
+    # collect round trip
+    full = np.concatenate(chunks)
+    assert np.array_equal(D.collect(), full)
+    # map + scalar broadcast
+    M = dja.dmap("sin", D)
+    assert np.allclose(M.collect(), np.sin(full), rtol=1e-15)
+    P = dja.elementwise_scalar("add", D, 1.0)
+    assert np.allclose(P.collect(), full + 1.0, rtol=0)
+    # distribute() scatter path
+    g = _global_f64(n, seed=42)
+    E = dja.distribute(g)
+    assert np.array_equal(E.collect(), g)
+    for d in (D, M, P, E):
+        d.close()
+
+
+def _scenario_routing(rank, world, dja):
+    """Mismatched-cuts transparent routing: ragged vs even layouts
+    (the ADVICE-r1 garbage-read scenario, now a correct gather path)."""
+    n = 1000
+    sizes = [n // world + (100 if r == 0 else 0) -
+             (100 if r == world - 1 else 0) for r in range(world)]
+    assert sum(sizes) == n
+    ga = _global_f64(n, 1)
+    gr = _global_f64(n, 2)
+    A = _slice_set(dja.DArray((n,), "f64"), ga)          # even cuts
+    R = dja.DArray.from_chunk_sizes(sizes, "f64")        # ragged cuts
+    _slice_set(R, gr)
+    assert not A.samedist(R)
+    # map2_ routes through gather (would read OOB before the fix)
+    Dst = dja.DArray((n,), "f64")
+    dja.map2_("add", Dst, A, R)
+    assert np.allclose(Dst.collect(), ga + gr, rtol=0)
+    # reversed destination layout (ragged dest)
+    Dst2 = dja.DArray.from_chunk_sizes(sizes, "f64")
+    dja.map2_("mul", Dst2, A, R)
+    assert np.allclose(Dst2.collect(), ga * gr, rtol=0)
+    # add_ / broadcast_fma / scalar / == / dot
+    A2 = A.copy()
+    dja.add_(A2, R, 2.0)
+    assert np.allclose(A2.collect(), ga + 2.0 * gr, rtol=0)
+    F = dja.DArray((n,), "f64")
+    dja.broadcast_fma(F, A, R, 0.25)
+    assert np.allclose(F.collect(), ga * gr + 0.25, rtol=0)
+    S = dja.DArray((n,), "f64")
+    dja.map2_scalar_("sub", S, R, 1.5)
+    assert np.allclose(S.collect(), gr - 1.5, rtol=0)
+    # == across layouts (same values -> True; perturbed -> False)
+    Rcopy = dja.DArray.from_chunk_sizes(sizes, "f64")
+    _slice_set(Rcopy, ga)
+    assert (A == Rcopy) is True
+    _slice_set(Rcopy, ga + 1e-9)
+    assert (A == Rcopy) is False
+    got = dja.ddot(A, R)
+    ref = float(np.dot(ga, gr))
+    assert abs(got - ref) <= 1e-12 * abs(ref)
+    # map_ with mismatched layouts
+    M = dja.DArray.from_chunk_sizes(sizes, "f64")
+    dja.map_("abs2", M, A)
+    assert np.allclose(M.collect(), ga * ga, rtol=0)
+    for d in (A, R, Dst, Dst2, A2, F, S, Rcopy, M):
+        d.close()
+
+
+def _scenario_matmul(rank, world, dja):
+    from oracle import ops as oops
+    m, kk, n = 24, 9, 16
+    ga = _global_f64(m * kk, 3).reshape((m, kk), order="F")
+    gb = _global_f64(kk * n, 4).reshape((kk, n), order="F")
+    # explicit 2x2 grids: J=K=2 so BOTH the b-slab all-to-all and the
+    # per-k partial exchange run (the cfg-4 dataflow shape)
+    A = _slice_set(dja.DArray((m, kk), "f64", (2, 2)), ga)
+    B = _slice_set(dja.DArray((kk, n), "f64", (2, 2)), gb)
+    C = dja.dmatmul(A, B)
+    ref = oops.oracle_matmul_blocked(ga, gb, A.cuts[0], A.cuts[1],
+                                     C.cuts[1])
+    assert np.allclose(C.collect(), ref, rtol=1e-12)
+    assert np.allclose(C.collect(), ga @ gb, rtol=1e-12)
+    # alpha/beta via dmul_
+    C0g = _global_f64(C.size, 5).reshape(C.dims, order="F")
+    C2 = _slice_set(dja.DArray(C.dims, "f64", C.dist), C0g)
+    dja.dmul_(C2, A, B, alpha=0.5, beta=2.0)
+    assert np.allclose(C2.collect(), 0.5 * (ga @ gb) + 2.0 * C0g,
+                       rtol=1e-12)
+    for d in (A, B, C, C2):
+        d.close()
+
+
+def _scenario_matmul_nooverlap(rank, world, dja):
+    """Single-group partial exchange (DA_MM_OVERLAP=0 fallback path)."""
+    os.environ["DA_MM_OVERLAP"] = "0"
+    try:
+        _scenario_matmul(rank, world, dja)
+    finally:
+        del os.environ["DA_MM_OVERLAP"]
+
+
+def _scenario_matmul_emptyk(rank, world, dja):
+    """k-dimension smaller than the J chunk count: zero-size k-cuts
+    (cuts1d pads with empty chunks) — the ADVICE-r1 IndexError/deadlock
+    scenario for partials."""
+    m, kk, n = 8, 1, 8
+    ga = _global_f64(m * kk, 6).reshape((m, kk), order="F")
+    gb = _global_f64(kk * n, 7).reshape((kk, n), order="F")
+    A = _slice_set(dja.DArray((m, kk), "f64", (2, 2)), ga)
+    assert any(hi == lo for (lo, hi) in
+               [A.idxs[c][1] for c in range(A.nchunks)]), \
+        "test setup: expected an empty k-cut"
+    B = _slice_set(dja.DArray((kk, n), "f64", (1, 2)), gb)
+    C = dja.dmatmul(A, B)
+    assert np.allclose(C.collect(), ga @ gb, rtol=1e-12)
+    A.close(); B.close(); C.close()
+
+
+def _scenario_matmul_b_outside(rank, world, dja):
+    """B owners outside A's process grid (linalg.jl:211-226 arbitrary-B
+    case): A lives on a 1x1 grid, B's second chunk on rank 1."""
+    m, kk, n = 8, 4, 6
+    ga = _global_f64(m * kk, 8).reshape((m, kk), order="F")
+    gb = _global_f64(kk * n, 9).reshape((kk, n), order="F")
+    A = _slice_set(dja.DArray((m, kk), "f64", (1, 1)), ga)
+    B = _slice_set(dja.DArray((kk, n), "f64", (2, 1)), gb)
+    C = dja.dmatmul(A, B)
+    assert np.allclose(C.collect(), ga @ gb, rtol=1e-12)
+    A.close(); B.close(); C.close()
+
+
+def _scenario_matmul_nonidentity_raises(rank, world, dja):
+    """A dims-reduction result (non-identity ranks) must be rejected
+    loudly, not silently mis-scheduled."""
+    from distributedarrays_jl_amd._ffi import DArrayError
+    g = _global_f64(64, 10).reshape((8, 8), order="F")
+    D = _slice_set(dja.DArray((8, 8), "f64", (2, 2)), g)
+    R = dja.dsum_dims(D, (0,))           # (1,8) on ranks [0, 2]
+    assert R.ranks != list(range(R.nchunks))
+    gb = _global_f64(8 * 4, 18).reshape((8, 4), order="F")
+    B = _slice_set(dja.DArray((8, 4), "f64", (2, 2)), gb)
+    try:
+        dja.dmatmul(R, B)
+        assert False, "expected DArrayError (non-identity ranks)"
+    except DArrayError:
+        pass
+    try:
+        dja.dmatvec(R, np.ones(8))
+        assert False, "expected DArrayError (non-identity ranks)"
+    except DArrayError:
+        pass
+    D.close(); R.close(); B.close()
+
+
+def _scenario_dims_reduce(rank, world, dja):
+    from oracle import ops as oops
+    dims, dist = (12, 10), (2, world // 2)
+    g = _global_f64(120, 11).reshape(dims, order="F")
+    D = _slice_set(dja.DArray(dims, "f64", dist), g)
+    for red in [(0,), (1,), (0, 1)]:
+        R = dja.dsum_dims(D, red)
+        ref = oops.oracle_reduce_dims(
+            "identity", "add",
+            [np.asfortranarray(g[tuple(slice(lo, hi) for lo, hi in ix)])
+             for ix in D.idxs], D.idxs, dims, set(red))
+        assert np.allclose(R.collect(), ref, rtol=1e-12), red
+        R.close()
+    # duplicate dims in mean: (0,0) == (0,)
+    M1 = dja.dmean_dims(D, (0, 0))
+    M2 = dja.dmean_dims(D, (0,))
+    assert np.allclose(M1.collect(), M2.collect(), rtol=0)
+    assert np.allclose(M1.collect(), g.mean(axis=0, keepdims=True),
+                       rtol=1e-12)
+    M1.close(); M2.close(); D.close()
+
+
+def _scenario_sort(rank, world, dja):
+    n = 5003
+    g = _global_f64(n, 12)
+    D = _slice_set(dja.DArray((n,), "f64"), g)
+    S = dja.dsort(D)
+    assert np.array_equal(S.collect(), np.sort(g))
+    # the sorted (ragged) result participates in aligned-or-routed ops:
+    # add_ with an even-cut DVector must route, not read OOB
+    E = _slice_set(dja.DArray((n,), "f64"), g)
+    dja.add_(E, S)
+    assert np.allclose(E.collect(), g + np.sort(g), rtol=0)
+    D.close(); S.close(); E.close()
+
+
+def _scenario_matvec(rank, world, dja):
+    m, kk = 12, 9
+    ga = _global_f64(m * kk, 13).reshape((m, kk), order="F")
+    x = _global_f64(kk, 14)
+    A = _slice_set(dja.DArray((m, kk), "f64", (2, world // 2)), ga)
+    y = dja.dmatvec(A, x, alpha=1.5)
+    assert np.allclose(y.collect(), 1.5 * (ga @ x), rtol=1e-12)
+    A.close(); y.close()
+
+
+def _scenario_halo(rank, world, dja):
+    """dgetindex + dtranspose + diag scaling, multi-rank."""
+    m, n = 10, 8
+    g = _global_f64(m * n, 15).reshape((m, n), order="F")
+    D = _slice_set(dja.DArray((m, n), "f64"), g)
+    box = ((1, 7), (2, 8))
+    got = dja.dgetindex(D, *box)
+    assert np.array_equal(got, g[1:7, 2:8])
+    T = dja.dtranspose(D)
+    assert np.array_equal(T.collect(), np.asfortranarray(g.T))
+    dvec = _global_f64(m, 16)
+    dja.ddiag_lmul(dvec, D)
+    assert np.allclose(D.collect(), dvec[:, None] * g, rtol=0)
+    D.close(); T.close()
+
+
+def _scenario_spmd(rank, world, dja):
+    from distributedarrays_jl_amd import spmd
+    # bcast
+    arr = np.arange(10, dtype=np.float64) * (rank + 1)
+    got = spmd.bcast_host(arr, root=0)
+    assert np.array_equal(got, np.arange(10, dtype=np.float64))
+    # scatter: root splits, each rank gets its part
+    parts = [np.full(4, float(r), dtype=np.float64)
+             for r in range(world)]
+    mine = spmd.scatter_host(parts, root=0)
+    assert np.array_equal(mine, np.full(4, float(rank)))
+    # gather
+    contrib = np.full(3, float(rank) + 0.5, dtype=np.float64)
+    res = spmd.gather_host(contrib, root=0)
+    if rank == 0:
+        for r in range(world):
+            assert np.array_equal(res[r], np.full(3, float(r) + 0.5))
+    else:
+        assert res is None
+    spmd.barrier()
+    # device sendto/recvfrom ring (rank r -> r+1 mod world)
+    import ctypes
+    from distributedarrays_jl_amd.ops import _Buf
+    from distributedarrays_jl_amd._ffi import check, lib
+    src = np.full(8, float(rank), dtype=np.float64)
+    sb, rb = _Buf(64), _Buf(64)
+    check(lib.da_h2d(sb.p, src.ctypes.data_as(ctypes.c_void_p), 64))
+    spmd.sendrecv(sb.p, (rank + 1) % world, rb.p,
+                  (rank - 1) % world, 64)
+    out = np.empty(8, dtype=np.float64)
+    check(lib.da_d2h(rb.p, out.ctypes.data_as(ctypes.c_void_p), 64))
+    assert np.array_equal(out, np.full(8, float((rank - 1) % world)))
+    sb.free(); rb.free()
+
+
+def _scenario_scalar_index(rank, world, dja):
+    n = 40
+    g = _global_f64(n, 17)
+    D = _slice_set(dja.DArray((n,), "f64"), g)
+    assert D.getindex(n - 1) == g[n - 1]
+    D.setindex(3.25, 2)
+    assert D.getindex(2) == 3.25
+    D.close()
+
+
+# ------------------------------------------------------------- test entry
+SCENARIOS_W2 = ["basic", "routing", "matmul_b_outside", "sort", "spmd",
+                "scalar_index"]
+SCENARIOS_W4 = ["basic", "routing", "matmul", "matmul_nooverlap",
+                "matmul_emptyk", "matmul_nonidentity_raises",
+                "dims_reduce", "sort", "matvec", "halo", "spmd"]
+
+
+@pytest.mark.timeout(420)
+@pytest.mark.parametrize("name", SCENARIOS_W2)
+def test_world2(tmp_path, name):
+    _spawn(tmp_path, 2, name)
+
+
+@pytest.mark.timeout(420)
+@pytest.mark.parametrize("name", SCENARIOS_W4)
+def test_world4(tmp_path, name):
+    _spawn(tmp_path, 4, name)
