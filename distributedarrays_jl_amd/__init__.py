@@ -16,7 +16,7 @@ imported here.
 """
 
 from ._ffi import DArrayError
-from . import comm, geometry, plan, spmd
+from . import comm, geometry, plan, spmd, expr
 from .darray import (DArray, dzeros, dones, dfill, drand, drandn,
                      distribute, localpart, localindices, d_closeall,
                      bytes_in_use, ddata, dgather, locate, allowscalar)
@@ -29,7 +29,7 @@ from .ops import (map_, dmap, map2_, elementwise, map2_scalar_, elementwise_scal
                   dgetindex, dmul_)
 
 __all__ = [
-    "DArray", "DArrayError", "comm", "geometry", "plan", "spmd",
+    "DArray", "DArrayError", "comm", "geometry", "plan", "spmd", "expr",
     "dzeros", "dones", "dfill", "drand", "drandn", "distribute",
     "localpart", "localindices", "d_closeall", "bytes_in_use",
     "ddata", "dgather", "locate", "allowscalar",

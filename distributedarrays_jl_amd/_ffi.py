@@ -53,6 +53,8 @@ _sigs = {
     "da_map": ([i32, ptr, ptr, u64, i32], i32),
     "da_map2": ([i32, ptr, ptr, ptr, u64, i32], i32),
     "da_bcast_fma": ([ptr, ptr, ptr, f64, u64, i32], i32),
+    "da_expr": ([ptr, i32, ptr, ptr, i32, ptr, ptr, i32, ptr, i32,
+                 u64, i32], i32),
     "da_map2_scalar": ([i32, ptr, ptr, f64, i32, u64, i32], i32),
     "da_axpby": ([ptr, ptr, f64, f64, u64, i32], i32),
     "da_add": ([ptr, ptr, f64, u64, i32], i32),

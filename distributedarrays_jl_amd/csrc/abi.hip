@@ -303,6 +303,16 @@ int da_bcast_fma(void* d, const void* a, const void* b, double c,
     return launch_bcast_fma(d, a, b, c, n, dtype, st().stream);
 }
 
+int da_expr(const int32_t* prog, int prog_len, void* dst,
+            const uint64_t* dst_dims, int nd,
+            void* const* srcs, const uint64_t* src_strides, int nsrcs,
+            const double* consts, int nconsts, uint64_t n, int dtype) {
+    DA_REQUIRE_INIT();
+    return launch_expr(prog, prog_len, dst, dst_dims, nd, srcs,
+                       src_strides, nsrcs, consts, nconsts, n, dtype,
+                       st().stream);
+}
+
 int da_axpby(void* y, const void* x, double alpha, double beta,
              uint64_t n, int dtype) {
     DA_REQUIRE_INIT();

@@ -96,6 +96,11 @@ int launch_bcast_fma(void* d, const void* a, const void* b, double c,
                      uint64_t n, int dtype, hipStream_t s);
 int launch_map2_scalar(int opcode, void* dst, const void* src, double c,
                        int rev, uint64_t n, int dtype, hipStream_t s);
+int launch_expr(const int32_t* prog, int plen, void* dst,
+                const uint64_t* dst_dims, int nd,
+                void* const* srcs, const uint64_t* src_strides, int nsrcs,
+                const double* consts, int nconsts,
+                uint64_t n, int dtype, hipStream_t s);
 int launch_transpose(void* dst, const void* src, uint64_t m, uint64_t n,
                      int dtype, hipStream_t s);
 int launch_diag_scale(void* a, uint64_t m, uint64_t n, const void* diag,

@@ -112,6 +112,30 @@ int da_map2(int opcode, void* dst, const void* a, const void* b,
             uint64_t n, int dtype);                          /* elementwise +,-,.. mapreduce.jl:180-189 */
 int da_bcast_fma(void* d, const void* a, const void* b, double c,
                  uint64_t n, int dtype);                     /* D .= A .* B .+ c, broadcast.jl:65-85 */
+
+/* Fused broadcast composition — materializes an arbitrary Broadcasted
+ * tree in one pass (copyto!(localpart, bclocal(bc)), broadcast.jl:65-98;
+ * nested broadcast pinned at test/darray.jl:880-912).
+ *
+ * prog: postfix int32 program, kind = ins>>8, idx = ins&0xff:
+ *   kind 0 = unary da_mapop on stack top; kind 1 = push argument idx;
+ *   kind 2 = push constant idx; kind 3 = binary da_map2op (pops rhs
+ *   then lhs).  Stack depth <= DA_EXPR_MAXSTACK (validated).
+ * dst_dims/nd: destination LOCAL chunk shape (column-major), nd <= 4.
+ * src_strides: nsrcs x nd element strides (row-major per arg); a 0
+ *   stride expands a Julia-broadcast singleton dim.  NULL = every arg
+ *   dense over the destination chunk (fast flat path).
+ * Numerics are bit-identical to da_map/da_map2 chains of the same ops
+ * (shared scalar functor tables, -ffp-contract=off). */
+#define DA_EXPR_MAXLEN   40
+#define DA_EXPR_MAXARGS  6
+#define DA_EXPR_MAXCONSTS 6
+#define DA_EXPR_MAXND    4
+#define DA_EXPR_MAXSTACK 8
+int da_expr(const int32_t* prog, int prog_len, void* dst,
+            const uint64_t* dst_dims, int nd,
+            void* const* srcs, const uint64_t* src_strides, int nsrcs,
+            const double* consts, int nconsts, uint64_t n, int dtype);
 int da_map2_scalar(int opcode, void* dst, const void* src, double c,
                    int reverse, uint64_t n, int dtype);      /* D .+ 1 etc (scalar broadcast arg, broadcast.jl:124-133) */
 int da_axpby(void* y, const void* x, double alpha, double beta,

@@ -180,6 +180,36 @@ class FakeLib:
         _tv(d, n, dt)[:] = oops.oracle_bcast_fma(av, bv, c)
         return 0
 
+    def da_expr(self, prog, plen, dst, dst_dims, nd, srcs, src_strides,
+                nsrcs, consts, nconsts, n, dtype):
+        import oracle.expr as oexpr
+        dt = _NPDT[int(dtype)]
+        plen, nd = int(plen), int(nd)
+        nsrcs, nconsts, n = int(nsrcs), int(nconsts), int(n)
+        pv = ctypes.cast(prog, ctypes.POINTER(ctypes.c_int32))
+        program = [int(pv[i]) for i in range(plen)]
+        cv = ctypes.cast(consts, ctypes.POINTER(ctypes.c_double))
+        cc = [float(cv[i]) for i in range(nconsts)]
+        sv = ctypes.cast(srcs, ctypes.POINTER(ctypes.c_void_p))
+        if _addr(src_strides) == 0:
+            args = [_tv(sv[i], n, dt) for i in range(nsrcs)]
+            out = oexpr.evaluate(program, args, cc, dt)
+            _tv(dst, n, dt)[:] = out.ravel()
+            return 0
+        dims = ctypes.cast(dst_dims, ctypes.POINTER(ctypes.c_uint64))
+        shape = tuple(int(dims[d]) for d in range(nd))
+        stv = ctypes.cast(src_strides, ctypes.POINTER(ctypes.c_uint64))
+        args = []
+        for i in range(nsrcs):
+            ss = [int(stv[i * nd + d]) for d in range(nd)]
+            base = _tv(sv[i], 1, dt)   # element 0; as_strided walks on
+            args.append(np.lib.stride_tricks.as_strided(
+                base, shape=shape,
+                strides=tuple(s * dt.itemsize for s in ss)))
+        out = oexpr.evaluate(program, args, cc, dt)
+        _tv(dst, n, dt)[:] = np.asfortranarray(out).ravel(order="F")
+        return 0
+
     def da_axpby(self, y, x, alpha, beta, n, dtype):
         dt = _NPDT[int(dtype)]
         yv, xv = _tv(y, n, dt), _tv(x, n, dt)
@@ -429,9 +459,9 @@ class FakeLib:
 def install():
     """Swap the package's ABI binding for a FakeLib — explicit,
     test-side-only.  Returns the instance."""
-    from distributedarrays_jl_amd import _ffi, ops, darray, spmd
+    from distributedarrays_jl_amd import _ffi, ops, darray, spmd, expr
     fake = FakeLib()
     _ffi.lib = fake
-    for mod in (ops, darray, spmd):
+    for mod in (ops, darray, spmd, expr):
         mod.lib = fake
     return fake

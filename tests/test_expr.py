@@ -1,0 +1,124 @@
+"""CPU tests of the broadcast-expression compiler (expr.py): postfix
+encoding, validation, and the oracle evaluator's agreement with plain
+numpy composition.  GPU parity of da_expr itself: test_gpu_expr.py."""
+import numpy as np
+import pytest
+
+from distributedarrays_jl_amd import expr as E
+from distributedarrays_jl_amd._ffi import DArrayError
+from distributedarrays_jl_amd._opcodes import MAP_OP, MAP2_OP
+import oracle.expr as oexpr
+from oracle import philox
+
+
+class FakeD:
+    """Metadata-only stand-in for compile tests (no GPU)."""
+
+    def __init__(self, dims, dtype="f64"):
+        self.dims = dims
+        self.dtype = dtype
+        self.lidx = None
+
+    @property
+    def ndims(self):
+        return len(self.dims)
+
+
+def test_encoding_pinned():
+    a, b = FakeD((4,)), FakeD((4,))
+    e = E.sin(E.ref(a)) + E.ref(b) * 0.5
+    prog, args, consts = E.compile_expr(e)
+    assert prog == [
+        (1 << 8) | 0,                  # push a
+        (0 << 8) | MAP_OP["sin"],      # sin
+        (1 << 8) | 1,                  # push b
+        (2 << 8) | 0,                  # push 0.5
+        (3 << 8) | MAP2_OP["mul"],     # *
+        (3 << 8) | MAP2_OP["add"],     # +
+    ]
+    assert len(args) == 2 and consts == [0.5]
+
+
+def test_leaf_dedup():
+    a = FakeD((4,))
+    e = E.ref(a) * E.ref(a) + E.ref(a)   # same object -> one arg slot
+    prog, args, consts = E.compile_expr(e)
+    assert len(args) == 1
+    assert sum(1 for p in prog if p >> 8 == 1) == 3
+
+
+def test_operator_sugar():
+    a = FakeD((4,))
+    forms = [
+        (-E.ref(a), "neg"),
+        (abs(E.ref(a)), "abs"),
+        (2.0 - E.ref(a), "sub"),
+        (2.0 / E.ref(a), "div"),
+        (E.ref(a) ** 2.0, "pow"),
+        (E.ref(a) % 3.0, "mod"),
+    ]
+    for e, op in forms:
+        prog, _, _ = E.compile_expr(e)
+        kinds = [(p >> 8, p & 0xFF) for p in prog]
+        table = MAP_OP if op in MAP_OP else MAP2_OP
+        kind = 0 if op in MAP_OP else 3
+        assert (kind, table[op]) in kinds, op
+
+
+def test_limits():
+    a = FakeD((4,))
+    with pytest.raises(DArrayError):
+        E.compile_expr(sum((E.ref(FakeD((4,))) for _ in range(7)),
+                           E.ref(a)))      # > MAXARGS distinct arrays
+    e = E.ref(a)
+    for i in range(8):
+        e = e + float(i + 2)               # distinct consts
+    with pytest.raises(DArrayError):
+        E.compile_expr(e)
+
+
+def test_oracle_eval_matches_numpy():
+    A = philox.fill_uniform_f64(200, 1).reshape((20, 10), order="F")
+    B = philox.fill_uniform_f64(10, 2).reshape((1, 10), order="F")
+    C = philox.fill_uniform_f64(200, 3).reshape((20, 10), order="F")
+    fa, fb, fc = FakeD((20, 10)), FakeD((1, 10)), FakeD((20, 10))
+    e = E.ref(fa) - E.ref(fb) * E.sin(E.ref(fc))
+    prog, args, consts = E.compile_expr(e)
+    out = oexpr.evaluate(prog, [A, B, C], consts)
+    assert np.array_equal(out, A - B * np.sin(C))
+    # chain with consts and division
+    e2 = (E.ref(fa) + E.ref(fc)) / (abs(E.ref(fa)) + 1.0)
+    prog2, args2, consts2 = E.compile_expr(e2)
+    out2 = oexpr.evaluate(prog2, [A, C], consts2)
+    assert np.array_equal(out2, (A + C) / (np.abs(A) + 1.0))
+
+
+def test_oracle_tables_match_opcode_tables():
+    """oracle/expr.py's index->name lists must mirror _opcodes.py (which
+    tests/test_abi.py pins against the C header)."""
+    from distributedarrays_jl_amd._opcodes import MAP_OPS, MAP2_OPS
+    assert oexpr.MAP_NAMES == MAP_OPS
+    assert oexpr.MAP2_NAMES == MAP2_OPS
+
+
+def test_i64_validation():
+    a, d = FakeD((4,), "i64"), FakeD((4,), "i64")
+    prog, args, consts = E.compile_expr(E.sin(E.ref(a)))
+    with pytest.raises(DArrayError):
+        E._validate(d, prog, args, consts)
+    prog, args, consts = E.compile_expr(E.ref(a) + 2.5)
+    with pytest.raises(DArrayError):
+        E._validate(d, prog, args, consts)   # non-integer i64 const
+    prog, args, consts = E.compile_expr(abs(E.ref(a)) * 3.0)
+    E._validate(d, prog, args, consts)       # fine
+
+
+def test_shape_validation():
+    d = FakeD((8, 4))
+    bad = FakeD((8, 3))
+    prog, args, consts = E.compile_expr(E.ref(bad) + 1.0)
+    with pytest.raises(DArrayError):
+        E._validate(d, prog, args, consts)
+    ok = FakeD((1, 4))
+    prog, args, consts = E.compile_expr(E.ref(ok) + 1.0)
+    E._validate(d, prog, args, consts)

@@ -328,6 +328,33 @@ def _scenario_spmd(rank, world, dja):
     sb.free(); rb.free()
 
 
+def _scenario_expr(rank, world, dja):
+    """Fused broadcast composition multi-rank: aligned args, a
+    dims-expanded mean operand, and a mismatched-cuts operand in ONE
+    tree (broadcast.jl:65-98 localisation)."""
+    from distributedarrays_jl_amd import expr as E
+    nr, nc = 16, 4 * world
+    g = _global_f64(nr * nc, 20).reshape((nr, nc), order="F")
+    A = _slice_set(dja.DArray((nr, nc), "f64", (1, world)), g)
+    M = dja.dmean_dims(A, (0,))
+    D = dja.DArray((nr, nc), "f64", (1, world))
+    E.materialize_(D, E.ref(A) - E.ref(M))
+    ref = g - g.mean(axis=0, keepdims=True)
+    assert np.allclose(D.collect(), ref, rtol=1e-12)
+    # nested: g2 = a .- m .* sin.(c) (the pinned reference form)
+    G = dja.DArray((nr, nc), "f64", (1, world))
+    E.materialize_(G, E.ref(A) - E.ref(M) * E.sin(E.ref(D)))
+    ref2 = g - g.mean(axis=0, keepdims=True) * np.sin(ref)
+    assert np.allclose(G.collect(), ref2, rtol=1e-12)
+    # mismatched-cuts operand: row-split B against column-split dest
+    B = _slice_set(dja.DArray((nr, nc), "f64", (world, 1)), g)
+    H = dja.DArray((nr, nc), "f64", (1, world))
+    E.materialize_(H, E.ref(A) * E.ref(B) + 0.5)
+    assert np.allclose(H.collect(), g * g + 0.5, rtol=0)
+    for d in (A, M, D, G, B, H):
+        d.close()
+
+
 def _scenario_scalar_index(rank, world, dja):
     n = 40
     g = _global_f64(n, 17)
@@ -340,10 +367,10 @@ def _scenario_scalar_index(rank, world, dja):
 
 # ------------------------------------------------------------- test entry
 SCENARIOS_W2 = ["basic", "routing", "matmul_b_outside", "sort", "spmd",
-                "scalar_index"]
+                "scalar_index", "expr"]
 SCENARIOS_W4 = ["basic", "routing", "matmul", "matmul_nooverlap",
                 "matmul_emptyk", "matmul_nonidentity_raises",
-                "dims_reduce", "sort", "matvec", "halo", "spmd"]
+                "dims_reduce", "sort", "matvec", "halo", "spmd", "expr"]
 
 
 @pytest.mark.timeout(420)
