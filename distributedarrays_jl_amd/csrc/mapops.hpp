@@ -8,6 +8,7 @@
 // (sign/round/sinc/cosc/mod noted inline); compiled -ffp-contract=off.
 #pragma once
 #include "darray_hip.h"
+#include "fastmath.hpp"
 
 namespace da {
 
@@ -35,7 +36,7 @@ __device__ __forceinline__ T apply_map(int op, T x) {
     case DA_OP_INV: return one / x;
     case DA_OP_SQRT: return sqrt(x);
     case DA_OP_CBRT: return cbrt(x);
-    case DA_OP_EXP: return exp(x);
+    case DA_OP_EXP: return da_exp(x);
     case DA_OP_EXP2: return exp2(x);
     case DA_OP_EXP10: return pow((T)10, x);
     case DA_OP_EXPM1: return expm1(x);
@@ -43,8 +44,8 @@ __device__ __forceinline__ T apply_map(int op, T x) {
     case DA_OP_LOG2: return log2(x);
     case DA_OP_LOG10: return log10(x);
     case DA_OP_LOG1P: return log1p(x);
-    case DA_OP_SIN: return sin(x);
-    case DA_OP_COS: return cos(x);
+    case DA_OP_SIN: return da_sin(x);
+    case DA_OP_COS: return da_cos(x);
     case DA_OP_TAN: return tan(x);
     case DA_OP_ASIN: return asin(x);
     case DA_OP_ACOS: return acos(x);
@@ -65,8 +66,8 @@ __device__ __forceinline__ T apply_map(int op, T x) {
         return x != x ? x : (x > zero ? one : (x < zero ? -one : x));
     case DA_OP_DEG2RAD: return x * (T)(M_PI / 180.0);
     case DA_OP_RAD2DEG: return x * (T)(180.0 / M_PI);
-    case DA_OP_SEC: return one / cos(x);
-    case DA_OP_CSC: return one / sin(x);
+    case DA_OP_SEC: return one / da_cos(x);
+    case DA_OP_CSC: return one / da_sin(x);
     case DA_OP_COT: return one / tan(x);
     case DA_OP_ERF: return erf(x);
     case DA_OP_ERFC: return erfc(x);
@@ -80,8 +81,8 @@ __device__ __forceinline__ T apply_map(int op, T x) {
     case DA_OP_COSC:   // Julia cosc: d/dx sinc = cospi(x)/x - sinpi(x)/(pi x^2)
         return x == zero ? zero
                          : cospi(x) / x - sinpi(x) / ((T)M_PI * x * x);
-    case DA_OP_SIND: return sin(x * (T)(M_PI / 180.0));
-    case DA_OP_COSD: return cos(x * (T)(M_PI / 180.0));
+    case DA_OP_SIND: return da_sin(x * (T)(M_PI / 180.0));
+    case DA_OP_COSD: return da_cos(x * (T)(M_PI / 180.0));
     case DA_OP_TAND: return tan(x * (T)(M_PI / 180.0));
     case DA_OP_ASIND: return asin(x) * (T)(180.0 / M_PI);
     case DA_OP_ACOSD: return acos(x) * (T)(180.0 / M_PI);
