@@ -55,6 +55,8 @@ _sigs = {
     "da_bcast_fma": ([ptr, ptr, ptr, f64, u64, i32], i32),
     "da_expr": ([ptr, i32, ptr, ptr, i32, ptr, ptr, i32, ptr, i32,
                  u64, i32], i32),
+    "da_expr_jit_state": ([], i32),
+    "da_expr_jit_errstr": ([], ctypes.c_char_p),
     "da_map2_scalar": ([i32, ptr, ptr, f64, i32, u64, i32], i32),
     "da_axpby": ([ptr, ptr, f64, f64, u64, i32], i32),
     "da_add": ([ptr, ptr, f64, u64, i32], i32),

@@ -313,6 +313,12 @@ int da_expr(const int32_t* prog, int prog_len, void* dst,
                        st().stream);
 }
 
+/* 1 = JIT ready (unused yet), 2 = active, -1 = failed (interpreter);
+ * DA_EXPR_JIT=0 disables per call.  da_expr_jit_errstr() holds the
+ * last hipRTC diagnostic. */
+int da_expr_jit_state(void) { return expr_jit_state(); }
+const char* da_expr_jit_errstr(void) { return expr_jit_err(); }
+
 int da_axpby(void* y, const void* x, double alpha, double beta,
              uint64_t n, int dtype) {
     DA_REQUIRE_INIT();

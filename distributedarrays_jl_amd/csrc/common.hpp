@@ -101,6 +101,15 @@ int launch_expr(const int32_t* prog, int plen, void* dst,
                 void* const* srcs, const uint64_t* src_strides, int nsrcs,
                 const double* consts, int nconsts,
                 uint64_t n, int dtype, hipStream_t s);
+// hipRTC path (expr_jit.hip): 0 = launched, 1 = unavailable (use the
+// interpreter), <0 = error
+int launch_expr_jit(const int32_t* prog, int plen, void* dst,
+                    const uint64_t* dst_dims, int nd,
+                    void* const* srcs, const uint64_t* src_strides,
+                    int nsrcs, const double* consts, int nconsts,
+                    uint64_t n, int dtype, hipStream_t s);
+int expr_jit_state();
+const char* expr_jit_err();
 int launch_transpose(void* dst, const void* src, uint64_t m, uint64_t n,
                      int dtype, hipStream_t s);
 int launch_diag_scale(void* a, uint64_t m, uint64_t n, const void* diag,

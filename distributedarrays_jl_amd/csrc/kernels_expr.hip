@@ -213,6 +213,17 @@ int launch_expr(const int32_t* prog, int plen, void* dst,
     if (sp != 1) return set_err(-3, "da_expr: program leaves %d values", sp);
     if (n == 0) return 0;
 
+    // preferred path: hipRTC-compiled kernel for this exact program
+    // (register pressure of the expression only; numerics identical —
+    // generated code calls the same functor tables).  Falls back to
+    // the interpreter below when disabled or compilation fails.
+    {
+        int rc = launch_expr_jit(prog, plen, dst, dst_dims, nd, srcs,
+                                 src_strides, nsrcs, consts, nconsts, n,
+                                 dtype, s);
+        if (rc <= 0) return rc;
+    }
+
     ExprProg P;
     memset(&P, 0, sizeof(P));
     for (int i = 0; i < plen; ++i) P.ins[i] = prog[i];

@@ -241,31 +241,48 @@ class DArray:
 
     # ---- conversion (Array(::DArray); collect) ----
     def collect(self):
-        """Full array on every rank.  Cross-rank chunk movement is
-        control-plane (torch.distributed gloo when world>1), mirroring
-        the reference's remotecall gather (darray.jl:574ff)."""
+        """Full array on every rank (Array(::DArray), darray.jl:574ff).
+
+        Cross-rank chunk movement is a DEVICE path: one grouped RCCL
+        broadcast per chunk from its owner over xGMI (the reference's
+        remotecall gather re-expressed), then one D2H of each chunk
+        into the assembled host array.  Collective at nranks>1."""
         npdt = np.dtype(NUMPY_DTYPES[self.dtype])
-        local = self.localpart()
+        out = np.zeros(self.dims, dtype=npdt, order="F")
         if self.nranks == 1:
             # (the fast path must be gated on the WORLD size, not the
-            # chunk count: at nranks>1 every rank must join the gather
-            # below or the collective mismatches)
-            out = np.zeros(self.dims, dtype=npdt, order="F")
+            # chunk count: at nranks>1 every rank must join the
+            # broadcasts below or the collective mismatches)
             if self.lnumel:
                 sl = tuple(slice(lo, hi) for lo, hi in self.lidx)
-                out[sl] = local
+                out[sl] = self.localpart()
             return out
-        import torch.distributed as td
-        if not td.is_initialized():
-            raise _ffi.DArrayError(
-                "collect() with nranks>1 needs torch.distributed (gloo) "
-                "initialized for the control plane")
-        gathered = [None] * self.nranks
-        td.all_gather_object(gathered, local)
-        out = np.zeros(self.dims, dtype=npdt, order="F")
+        from . import ops
+        esz = DTYPE_SIZE[self.dtype]
+        bufs = []
         for c in range(self.nchunks):
+            nel = geometry.nelems(self.idxs[c])
+            buf = ops._Buf(max(nel, 1) * esz) if nel else None
+            if buf is not None and self.ranks[c] == self.rank:
+                check(lib.da_d2d(buf.p, self._ptr(), nel * esz))
+            bufs.append(buf)
+        check(lib.da_group_start())
+        for c, buf in enumerate(bufs):
+            if buf is not None:
+                check(lib.da_bcast(buf.p,
+                                   geometry.nelems(self.idxs[c]) * esz,
+                                   self.ranks[c]))
+        check(lib.da_group_end())
+        for c, buf in enumerate(bufs):
+            if buf is None:
+                continue
+            shape = geometry.shape_of(self.idxs[c])
+            host = np.empty(shape, dtype=npdt, order="F")
+            check(lib.da_d2h(buf.p, host.ctypes.data_as(ctypes.c_void_p),
+                             host.size * esz))
             sl = tuple(slice(lo, hi) for lo, hi in self.idxs[c])
-            out[sl] = gathered[self.ranks[c]]
+            out[sl] = host
+            buf.free()
         return out
 
     def getindex(self, *point):

@@ -136,6 +136,11 @@ int da_expr(const int32_t* prog, int prog_len, void* dst,
             const uint64_t* dst_dims, int nd,
             void* const* srcs, const uint64_t* src_strides, int nsrcs,
             const double* consts, int nconsts, uint64_t n, int dtype);
+/* da_expr compiles each distinct program to a dedicated kernel via
+ * hipRTC (cached per process; DA_EXPR_JIT=0 forces the interpreter).
+ * state: 1 ready, 2 active, -1 hipRTC failed (interpreter fallback). */
+int da_expr_jit_state(void);
+const char* da_expr_jit_errstr(void);
 int da_map2_scalar(int opcode, void* dst, const void* src, double c,
                    int reverse, uint64_t n, int dtype);      /* D .+ 1 etc (scalar broadcast arg, broadcast.jl:124-133) */
 int da_axpby(void* y, const void* x, double alpha, double beta,

@@ -210,6 +210,12 @@ class FakeLib:
         _tv(dst, n, dt)[:] = np.asfortranarray(out).ravel(order="F")
         return 0
 
+    def da_expr_jit_state(self):
+        return 0      # no JIT in the fake (numpy evaluator only)
+
+    def da_expr_jit_errstr(self):
+        return b""
+
     def da_axpby(self, y, x, alpha, beta, n, dtype):
         dt = _NPDT[int(dtype)]
         yv, xv = _tv(y, n, dt), _tv(x, n, dt)

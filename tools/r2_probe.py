@@ -104,6 +104,10 @@ def main():
         t = bench(lambda: E.materialize_(D, e3))
         out("expr_chain5", t * 1e3, n * 24 / t / 1e9)
         A.close(); B.close(); D.close(); T_.close()
+        print(json.dumps({"leg": "expr_jit_state",
+                          "state": int(lib.da_expr_jit_state()),
+                          "err": (lib.da_expr_jit_errstr() or b"")
+                          .decode()[:200]}), flush=True)
         # strided: x - mean(x, dims=1) on 16384^2 (2 GiB)
         m = 16384
         X = dja.drand((m, m), "f64")
