@@ -123,7 +123,7 @@ std::string gen_source(const int32_t* prog, int plen, int dtype, int nd,
         "#include \"mapops.hpp\"\n";
     s += JARGS_DECL;
     s += gen_eval(prog, plen, tname, i64, strided);
-    char buf[512];
+    char buf[4096];
     if (!strided) {
         snprintf(buf, sizeof(buf),
             "extern \"C\" __global__ void ejit(JArgs a) {\n"
@@ -280,3 +280,47 @@ int expr_jit_state() { return g_jit_state; }
 const char* expr_jit_err() { return g_jit_err; }
 
 } // namespace da
+
+/* test-only: generate + hiprtc-compile a program WITHOUT loading it
+ * (no GPU, no da_init needed — hipRTC is a pure compiler).  Lets the
+ * CPU suite pin the codegen.  Returns 0 ok, -1 compile failure (see
+ * da_expr_jit_errstr); src_out receives the generated source. */
+extern "C" int dbg_expr_jit_compile(const int32_t* prog, int plen,
+                                    int dtype, int nd, int nsrcs,
+                                    int strided, char* src_out,
+                                    int src_len) {
+    using namespace da;
+    std::string src = gen_source(prog, plen, dtype, nd, nsrcs,
+                                 strided != 0);
+    if (src_out && src_len > 0) {
+        strncpy(src_out, src.c_str(), src_len - 1);
+        src_out[src_len - 1] = 0;
+    }
+    const char* hdr_names[] = {"stdint.h", "darray_hip.h", "mapops.hpp",
+                               "fastmath.hpp"};
+    const char* stdint_stub =
+        "#pragma once\n"
+        "typedef __INT8_TYPE__ int8_t;\ntypedef __UINT8_TYPE__ uint8_t;\n"
+        "typedef __INT16_TYPE__ int16_t;\ntypedef __UINT16_TYPE__ uint16_t;\n"
+        "typedef __INT32_TYPE__ int32_t;\ntypedef __UINT32_TYPE__ uint32_t;\n"
+        "typedef __INT64_TYPE__ int64_t;\ntypedef __UINT64_TYPE__ uint64_t;\n";
+    const char* hdrs[] = {stdint_stub, embedded_darray_hip_h,
+                          embedded_mapops_hpp, embedded_fastmath_hpp};
+    hiprtcProgram p;
+    if (hiprtcCreateProgram(&p, src.c_str(), "ejit.cu", 4, hdrs,
+                            hdr_names) != HIPRTC_SUCCESS)
+        return -1;
+    const char* opts[] = {"--offload-arch=gfx950", "-O3", "-std=c++17",
+                          "-ffp-contract=off"};
+    hiprtcResult rc = hiprtcCompileProgram(p, 4, opts);
+    if (rc != HIPRTC_SUCCESS) {
+        size_t lsz = 0;
+        hiprtcGetProgramLogSize(p, &lsz);
+        std::string log(lsz, 0);
+        if (lsz) hiprtcGetProgramLog(p, &log[0]);
+        snprintf(g_jit_err, sizeof(g_jit_err), "hiprtc compile: %.400s",
+                 log.c_str());
+    }
+    hiprtcDestroyProgram(&p);
+    return rc == HIPRTC_SUCCESS ? 0 : -1;
+}

@@ -715,6 +715,76 @@ def dmatvec(A, x, alpha=1.0):
     return y
 
 
+def dmatvec_adj(A, x, alpha=1.0):
+    """y = alpha * A' * x — mul!(y, adjoint(A), x) (linalg.jl:124-167):
+    rank (i,j) multiplies its block TRANSPOSED by the x-slice of its ROW
+    cut (xj = x[A.cuts[1][j]-range], linalg.jl:139), partial vectors of
+    y-block j travel to rank j (y.cuts == A.cuts[2], linalg.jl:134) and
+    accumulate ascending i.  Real dtypes: adjoint == transpose, so this
+    also serves mul!(y, transpose(A), x) (linalg.jl:169 analog).
+
+    The local A_loc'*xj is one da_gemm_f64 with the vector as the 1-row
+    left operand: C(1 x kloc) = x_row(1 x mloc) * A_loc(mloc x kloc)."""
+    import numpy as np
+    if A.dtype != "f64" or A.ndims != 2:
+        raise DArrayError("dmatvec_adj: 2-D f64 only")
+    if A.ranks != list(range(A.nchunks)):
+        raise DArrayError(
+            "dmatvec_adj: A must have identity chunk->rank mapping")
+    x = np.ascontiguousarray(np.asarray(x, dtype=np.float64))
+    if x.shape != (A.dims[0],):
+        raise DArrayError("dmatvec_adj: x length %d != %d"
+                          % (x.shape[0], A.dims[0]))
+    I, J = A.dist
+    y = DArray((A.dims[1],), "f64", (J,))
+    y.fill_(0.0)
+    r = A.rank
+    esz = 8
+    partial = None
+    kloc = 0
+    if A.lchunk is not None and A.lnumel:
+        ilo, ihi = A.lidx[0]
+        xi = x[ilo:ihi]
+        xbuf = _Buf(max(xi.size, 1) * esz)
+        check(lib.da_h2d(xbuf.p, xi.ctypes.data_as(ctypes.c_void_p),
+                         xi.size * esz))
+        mloc, kloc = A.lshape
+        partial = _Buf(max(kloc, 1) * esz)
+        check(lib.da_gemm_f64(partial.p, xbuf.p, A._ptr(),
+                              1, kloc, mloc, 1, mloc, 1, 1.0, 0.0))
+        xbuf.free()
+
+    sends, recvs = [], {}
+    if A.lchunk is not None and A.lnumel:
+        i, j = r % I, r // I
+        if j != r and kloc:          # my partial's owner is rank j
+            sends.append((j, partial))
+    if r < J and y.lnumel:
+        for i in range(I):
+            src = i + I * r
+            if src != r:
+                recvs[src] = _Buf(max(y.lnumel, 1) * esz)
+    if sends or recvs:
+        check(lib.da_group_start())
+        for dst, buf in sends:
+            check(lib.da_send(buf.p, kloc * esz, dst))
+        for src, buf in recvs.items():
+            check(lib.da_recv(buf.p, y.lnumel * esz, src))
+        check(lib.da_group_end())
+    if r < J and y.lnumel:
+        for i in range(I):
+            src = i + I * r
+            buf = partial if src == r else recvs[src]
+            check(lib.da_add(y._ptr(), buf.p, float(alpha), y.lnumel,
+                             DTYPES["f64"]))
+    check(lib.da_synchronize())
+    if partial is not None:
+        partial.free()
+    for buf in recvs.values():
+        buf.free()
+    return y
+
+
 # ------------------------------------------------- makelocal halo gather
 def gather_box(A, boxes_all):
     """Collective makelocal (darray.jl:351-368): every rank passes the
