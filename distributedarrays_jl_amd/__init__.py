@@ -24,7 +24,8 @@ from .ops import (map_, dmap, map2_, elementwise, map2_scalar_, elementwise_scal
                   add_, scale_, mapreduce, dsum, dprod, dmaximum, dminimum,
                   dextrema, dmean, dcount, dall, dany, ddot, dnorm, dmatmul, dreduce_dims,
                   dsum_dims, dprod_dims, dmaximum_dims, dminimum_dims,
-                  dmean_dims, dmatvec, gather_box, map_general,
+                  dmean_dims, dmatvec, dmatvec_adj, gather_box, map_general,
+                  map2_general,
                   broadcast_fma_general, dsort, dtranspose, ddiag_lmul, ddiag_rmul,
                   dgetindex, dmul_)
 
@@ -37,7 +38,8 @@ __all__ = [
     "add_", "scale_", "mapreduce", "dsum", "dprod", "dmaximum",
     "dminimum", "dextrema", "dmean", "dcount", "dall", "dany", "ddot", "dnorm", "dmatmul",
     "dreduce_dims", "dsum_dims", "dprod_dims", "dmaximum_dims",
-    "dminimum_dims", "dmean_dims", "dmatvec",
+    "dminimum_dims", "dmean_dims", "dmatvec", "dmatvec_adj",
+    "map2_general",
     "gather_box", "map_general", "broadcast_fma_general", "dsort",
     "dtranspose", "ddiag_lmul", "ddiag_rmul", "dgetindex", "dmul_",
 ]

@@ -92,7 +92,13 @@ def test_gpu_matvec():
     y.close()
     y2 = dja.dmatvec(dA, x, alpha=2.0)
     assert np.allclose(y2.collect(), 2.0 * (A @ x), rtol=1e-12)
-    y2.close(); dA.close()
+    y2.close()
+    # adjoint form (linalg.jl:124-167): y = alpha * A' * x
+    xr = philox.fill_uniform_f64(m, 13)
+    ya = dja.dmatvec_adj(dA, xr, alpha=1.5)
+    assert ya.dims == (k,)
+    assert np.allclose(ya.collect(), 1.5 * (A.T @ xr), rtol=1e-12)
+    ya.close(); dA.close()
 
 
 @pytest.mark.gpu

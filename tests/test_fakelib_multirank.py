@@ -274,7 +274,11 @@ def _scenario_matvec(rank, world, dja):
     A = _slice_set(dja.DArray((m, kk), "f64", (2, world // 2)), ga)
     y = dja.dmatvec(A, x, alpha=1.5)
     assert np.allclose(y.collect(), 1.5 * (ga @ x), rtol=1e-12)
-    A.close(); y.close()
+    # adjoint: y2 = alpha * A' * x2 (linalg.jl:124-167)
+    x2 = _global_f64(m, 19)
+    y2 = dja.dmatvec_adj(A, x2, alpha=0.5)
+    assert np.allclose(y2.collect(), 0.5 * (ga.T @ x2), rtol=1e-12)
+    A.close(); y.close(); y2.close()
 
 
 def _scenario_halo(rank, world, dja):
