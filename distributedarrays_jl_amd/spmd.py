@@ -4,17 +4,66 @@ scatter/gather over per-worker RemoteChannels).  Here the payloads are
 device buffers and the transport is RCCL over xGMI; the `*_host`
 convenience forms stage numpy arrays through HBM.
 
-The reference's context machinery (spmd.jl:16-60) keyed channels per
-run; with RCCL the communicator IS the context, so contexts collapse to
-the single per-process communicator (SPMD's collectives are
-subsumed by RCCL)."""
+Context semantics (spmd.jl:16-60; concurrent runs pinned at
+test/spmd.jl:108-195): the reference keys per-worker RemoteChannels by
+context so concurrent @async SPMD runs cannot cross-match messages.
+Here a rank is an OS process issuing RCCL ops in program order on one
+communicator, and RCCL matches point-to-point ops per peer in issue
+order — so concurrent contexts are supported under the SPMD ordering
+contract: every rank must issue the interleaved contexts' operations
+in the SAME global order (which any deterministic SPMD program does,
+including the reference's own concurrent-runs test, whose runs are
+spawned in a fixed order).  Context objects carry the reference's
+context-local storage (test/spmd.jl:154-195) and an epoch guard that
+raises on out-of-order reuse instead of deadlocking."""
 import ctypes
+import itertools
 
 import numpy as np
 
 from . import comm
 from ._ffi import check, lib, DArrayError
 from .ops import _Buf
+
+_ctx_counter = itertools.count()
+_ctx_last_finished = [-1]
+
+
+class Context:
+    """An SPMD run context — context_local_storage() analog
+    (spmd.jl:16-60).  Creation order defines the serialization domain:
+    finish contexts (close_ctx) in creation order on every rank."""
+
+    def __init__(self):
+        self.id = next(_ctx_counter)
+        self.storage = {}
+        self.closed = False
+
+    def context_local_storage(self):
+        """Per-context dict (the reference returns a per-(context,
+        worker) Dict; here per-(context, rank))."""
+        if self.closed:
+            raise DArrayError("spmd context %d already closed" % self.id)
+        return self.storage
+
+    def close(self):
+        """close_ctx analog: frees storage and enforces creation-order
+        completion (the deterministic-order contract above)."""
+        if self.closed:
+            return
+        if self.id != _ctx_last_finished[0] + 1:
+            raise DArrayError(
+                "spmd contexts must complete in creation order "
+                "(closing %d after %d)" % (self.id,
+                                           _ctx_last_finished[0]))
+        _ctx_last_finished[0] = self.id
+        self.storage = {}
+        self.closed = True
+
+
+def context():
+    """spmd_context() analog."""
+    return Context()
 
 
 def _auto():

@@ -332,6 +332,55 @@ def _scenario_spmd(rank, world, dja):
     sb.free(); rb.free()
 
 
+def _scenario_spmd_contexts(rank, world, dja):
+    """Concurrent SPMD runs (test/spmd.jl:108-195 analog): 8 contexts
+    whose ring-exchange operations INTERLEAVE round-robin — pairing
+    holds because every rank issues the interleaved ops in the same
+    global order (the module's documented serialization contract) —
+    plus context-local storage isolation and the creation-order close
+    guard."""
+    import ctypes
+    from distributedarrays_jl_amd import spmd
+    from distributedarrays_jl_amd.ops import _Buf
+    from distributedarrays_jl_amd._ffi import check, lib, DArrayError
+    NRUNS = 8
+    ctxs = [spmd.context() for _ in range(NRUNS)]
+    for k, c in enumerate(ctxs):
+        c.context_local_storage()["val"] = rank * 100 + k
+    # one ring sendrecv step per context, interleaved round-robin
+    results = {}
+    for k, c in enumerate(ctxs):
+        src_arr = np.full(4, float(rank * NRUNS + k), dtype=np.float64)
+        sb, rb = _Buf(32), _Buf(32)
+        check(lib.da_h2d(sb.p, src_arr.ctypes.data_as(ctypes.c_void_p),
+                         32))
+        spmd.sendrecv(sb.p, (rank + 1) % world, rb.p,
+                      (rank - 1) % world, 32)
+        got = np.empty(4, dtype=np.float64)
+        check(lib.da_d2h(rb.p, got.ctypes.data_as(ctypes.c_void_p), 32))
+        results[k] = got
+        sb.free(); rb.free()
+    for k in range(NRUNS):
+        expect = float(((rank - 1) % world) * NRUNS + k)
+        assert np.array_equal(results[k], np.full(4, expect)), k
+    # storage is per-context, untouched by other runs
+    for k, c in enumerate(ctxs):
+        assert c.context_local_storage()["val"] == rank * 100 + k
+    # out-of-order close raises instead of deadlocking
+    try:
+        ctxs[3].close()
+        assert False, "expected DArrayError"
+    except DArrayError:
+        pass
+    for c in ctxs:
+        c.close()
+    try:
+        ctxs[0].context_local_storage()
+        assert False, "expected DArrayError on closed context"
+    except DArrayError:
+        pass
+
+
 def _scenario_expr(rank, world, dja):
     """Fused broadcast composition multi-rank: aligned args, a
     dims-expanded mean operand, and a mismatched-cuts operand in ONE
@@ -371,10 +420,11 @@ def _scenario_scalar_index(rank, world, dja):
 
 # ------------------------------------------------------------- test entry
 SCENARIOS_W2 = ["basic", "routing", "matmul_b_outside", "sort", "spmd",
-                "scalar_index", "expr"]
+                "scalar_index", "expr", "spmd_contexts"]
 SCENARIOS_W4 = ["basic", "routing", "matmul", "matmul_nooverlap",
                 "matmul_emptyk", "matmul_nonidentity_raises",
-                "dims_reduce", "sort", "matvec", "halo", "spmd", "expr"]
+                "dims_reduce", "sort", "matvec", "halo", "spmd", "expr",
+                "spmd_contexts"]
 
 
 @pytest.mark.timeout(420)
