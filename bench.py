@@ -126,10 +126,18 @@ def main():
     M.close()
     log("[bench] map!(sin): %.1f GB/s" % extra["map_sin_gbs"])
 
-    # ---------------- cfg3: Dd .= A .* B .+ c --------------------------
-    A3 = dja.DArray((n * world,), "f64", (world,)); A3.rand_()
-    B3 = dja.DArray((n * world,), "f64", (world,)); B3.rand_()
-    D3 = dja.dzeros((n * world,), "f64", (world,))
+    # ---------------- cfg3: D .= A .* B .+ c ---------------------------
+    # At world==4 this is EXACTLY BASELINE cfg3: 32768^2 Float64 on a
+    # 1-D column distribution (2 GiB/GPU); other worlds weak-scale the
+    # same bytes as a DVector.
+    if world == 4:
+        shp, dist3 = (32768, 32768), (1, 4)
+    else:
+        shp, dist3 = (n * world,), (world,)
+    A3 = dja.DArray(shp, "f64", dist3); A3.rand_()
+    B3 = dja.DArray(shp, "f64", dist3); B3.rand_()
+    D3 = dja.dzeros(shp, "f64", dist3)
+    n3 = A3.size // world
     for _ in range(W):
         dja.broadcast_fma(D3, A3, B3, 0.5)
     barrier()
@@ -138,10 +146,38 @@ def main():
         dja.broadcast_fma(D3, A3, B3, 0.5)
     barrier()
     t_bc = max_over_ranks(time.perf_counter() - t0)
-    extra["bcast_fma_gbs"] = world * n * 24.0 * K / t_bc / 1e9  # 2R+1W
+    extra["bcast_fma_gbs"] = world * n3 * 24.0 * K / t_bc / 1e9  # 2R+1W
+    extra["bcast_fma_config"] = "%r f64 dist %r" % (shp, dist3)
     for d in (A3, B3, D3):
         d.close()
     log("[bench] bcast fma: %.1f GB/s" % extra["bcast_fma_gbs"])
+
+    # ---------------- fused broadcast composition (da_expr JIT) --------
+    # D .= sin.(A) .+ B .* c in ONE kernel (nested broadcast,
+    # test/darray.jl:880-912 analog) — algorithmic 24 B/elem.
+    try:
+        from distributedarrays_jl_amd import expr as E
+        Ax = dja.DArray((n * world,), "f64", (world,)); Ax.rand_()
+        Bx = dja.DArray((n * world,), "f64", (world,)); Bx.rand_()
+        Dx = dja.dzeros((n * world,), "f64", (world,))
+        ex = E.sin(E.ref(Ax)) + E.ref(Bx) * 0.5
+        for _ in range(W):
+            E.materialize_(Dx, ex)
+        barrier()
+        t0 = time.perf_counter()
+        for _ in range(K):
+            E.materialize_(Dx, ex)
+        barrier()
+        t_ex = max_over_ranks(time.perf_counter() - t0)
+        extra["expr_fused_gbs"] = world * n * 24.0 * K / t_ex / 1e9
+        extra["expr_jit_state"] = int(lib.da_expr_jit_state())
+        for d in (Ax, Bx, Dx):
+            d.close()
+        log("[bench] expr sin-fused: %.1f GB/s (jit state %d)"
+            % (extra["expr_fused_gbs"], extra["expr_jit_state"]))
+    except Exception as e:
+        extra["expr_error"] = repr(e)[:200]
+        log("[bench] expr leg failed: %r" % (e,))
 
     # ---------------- cfg5 (N==8): mapreduce(abs2,+,f32 2^31) ----------
     if world == 8:
