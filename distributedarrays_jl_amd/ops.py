@@ -1001,8 +1001,6 @@ def dsort(d, samples_per_rank=64):
     import numpy as np
     if d.ndims != 1:
         raise DArrayError("dsort: DVector only")
-    if d.dtype == "f32":
-        raise DArrayError("dsort: f64/i64 only (round 1)")
     esz = DTYPE_SIZE[d.dtype]
     npdt = np.dtype(NUMPY_DTYPES[d.dtype])
     P = d.nranks

@@ -265,6 +265,13 @@ def _scenario_sort(rank, world, dja):
     dja.add_(E, S)
     assert np.allclose(E.collect(), g + np.sort(g), rtol=0)
     D.close(); S.close(); E.close()
+    # f32 path
+    from oracle import philox
+    g32 = philox.fill_uniform_f32(2001, 40)
+    D32 = _slice_set(dja.DArray((2001,), "f32"), g32)
+    S32 = dja.dsort(D32)
+    assert np.array_equal(S32.collect(), np.sort(g32, kind="stable"))
+    D32.close(); S32.close()
 
 
 def _scenario_matvec(rank, world, dja):

@@ -80,3 +80,14 @@ def test_gpu_dsort_i64():
         r = dja.dsort(d)
         assert np.array_equal(r.localpart(), np.sort(x))
         r.close(); d.close()
+
+
+@pytest.mark.gpu
+def test_gpu_dsort_f32():
+    import distributedarrays_jl_amd as dja
+    dja.comm.init()
+    x = philox.fill_uniform_f32(200003, seed=11)
+    d = dja.distribute(x)
+    r = dja.dsort(d)
+    assert np.array_equal(r.localpart(), np.sort(x, kind="stable"))
+    r.close(); d.close()
