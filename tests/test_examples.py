@@ -30,3 +30,17 @@ def test_game_of_life():
     ok, pop = run(128, 8)
     assert ok, "life diverged from numpy reference"
     assert pop > 0
+
+
+def test_zscore():
+    import numpy as np
+    from examples.zscore import zscore
+    X, mu, sig, Z = zscore(1024, 128)
+    hX = X.collect()
+    hZ = Z.collect()
+    ref = hX - hX.mean(axis=0, keepdims=True)
+    ref = ref / np.sqrt((hX * hX).mean(axis=0, keepdims=True)
+                        - hX.mean(axis=0, keepdims=True) ** 2)
+    assert np.max(np.abs(hZ - ref)) < 1e-11
+    for d in (X, mu, sig, Z):
+        d.close()
