@@ -287,29 +287,36 @@ static int do_reduce(int mapop, int redop, const T* src, uint64_t n,
 // coalesced across inner.  Variant B (inner < 64): one 64-lane wave per
 // (i,o), lanes stride the axis (coalesced along the axis), then a
 // wavefront shuffle tree.
-template <typename T>
+// MOP/ROP >= 0 are compile-time op constants (the switches fold, the
+// same 2x-recovery trick as map_fixed_kernel); -1 defers to the runtime
+// arguments (long-tail combos).
+template <typename T, int MOP, int ROP>
 __global__ void reduce_dims_threads(int mapop, int redop,
                                     const T* __restrict__ src,
                                     uint64_t inner, uint64_t axis,
                                     uint64_t outer, T* __restrict__ dst) {
+    const int mop_ = MOP >= 0 ? MOP : mapop;
+    const int rop_ = ROP >= 0 ? ROP : redop;
     uint64_t t = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
     uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
     uint64_t total = inner * outer;
     for (uint64_t e = t; e < total; e += stride) {
         uint64_t i = e % inner, o = e / inner;
         const T* p = src + i + o * inner * axis;
-        T acc = RedIdent<T>::get(redop);
+        T acc = RedIdent<T>::get(rop_);
         for (uint64_t a = 0; a < axis; ++a)
-            acc = red_comb(redop, acc, mapf(mapop, p[a * inner]));
+            acc = red_comb(rop_, acc, mapf(mop_, p[a * inner]));
         dst[i + o * inner] = acc;
     }
 }
 
-template <typename T>
+template <typename T, int MOP, int ROP>
 __global__ void reduce_dims_waves(int mapop, int redop,
                                   const T* __restrict__ src,
                                   uint64_t inner, uint64_t axis,
                                   uint64_t outer, T* __restrict__ dst) {
+    const int mop_ = MOP >= 0 ? MOP : mapop;
+    const int rop_ = ROP >= 0 ? ROP : redop;
     uint64_t wid = ((uint64_t)blockIdx.x * blockDim.x + threadIdx.x) / 64;
     uint64_t nw = ((uint64_t)gridDim.x * blockDim.x) / 64;
     int lane = threadIdx.x & 63;
@@ -317,12 +324,12 @@ __global__ void reduce_dims_waves(int mapop, int redop,
     for (uint64_t e = wid; e < total; e += nw) {
         uint64_t i = e % inner, o = e / inner;
         const T* p = src + i + o * inner * axis;
-        T acc = RedIdent<T>::get(redop);
+        T acc = RedIdent<T>::get(rop_);
         for (uint64_t a = lane; a < axis; a += 64)
-            acc = red_comb(redop, acc, mapf(mapop, p[a * inner]));
+            acc = red_comb(rop_, acc, mapf(mop_, p[a * inner]));
 #pragma unroll
         for (int off = 32; off > 0; off >>= 1)
-            acc = red_comb(redop, acc, (T)__shfl_down(acc, off, 64));
+            acc = red_comb(rop_, acc, (T)__shfl_down(acc, off, 64));
         if (lane == 0) dst[i + o * inner] = acc;
     }
 }
@@ -331,19 +338,21 @@ __global__ void reduce_dims_waves(int mapop, int redop,
 // over a long axis (e.g. sum(D, dims=2) of a square matrix), where the
 // thread/wave variants leave the chip underfilled (measured 137 GB/s vs
 // 3+ TB/s; profiles/r01_kernel_stats.md).
-template <typename T>
+template <typename T, int MOP, int ROP>
 __global__ void reduce_dims_blocks(int mapop, int redop,
                                    const T* __restrict__ src,
                                    uint64_t inner, uint64_t axis,
                                    uint64_t outer, T* __restrict__ dst) {
+    const int mop_ = MOP >= 0 ? MOP : mapop;
+    const int rop_ = ROP >= 0 ? ROP : redop;
     uint64_t total = inner * outer;
     for (uint64_t e = blockIdx.x; e < total; e += gridDim.x) {
         uint64_t i = e % inner, o = e / inner;
         const T* p = src + i + o * inner * axis;
-        T acc = RedIdent<T>::get(redop);
+        T acc = RedIdent<T>::get(rop_);
         for (uint64_t a = threadIdx.x; a < axis; a += blockDim.x)
-            acc = red_comb(redop, acc, mapf(mapop, p[a * inner]));
-        acc = block_reduce(redop, acc);
+            acc = red_comb(rop_, acc, mapf(mop_, p[a * inner]));
+        acc = block_reduce(rop_, acc);
         if (threadIdx.x == 0) dst[i + o * inner] = acc;
         __syncthreads();   // LDS in block_reduce reused next iteration
     }
@@ -354,20 +363,22 @@ __global__ void reduce_dims_blocks(int mapop, int redop,
 // its slice with fully coalesced reads (consecutive threads =
 // consecutive rows), producing NB partial rows; a second kernel folds
 // the NB partials per row.
-template <typename T>
+template <typename T, int MOP, int ROP>
 __global__ void reduce_dims_slices1(int mapop, int redop,
                                     const T* __restrict__ src,
                                     uint64_t inner, uint64_t axis,
                                     int nb, T* __restrict__ partials) {
+    const int mop_ = MOP >= 0 ? MOP : mapop;
+    const int rop_ = ROP >= 0 ? ROP : redop;
     uint64_t o = blockIdx.z;
     const T* base = src + o * inner * axis;
     uint64_t row = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
     if (row >= inner) return;
     uint64_t a0 = axis * blockIdx.y / nb;
     uint64_t a1 = axis * (blockIdx.y + 1) / nb;
-    T acc = RedIdent<T>::get(redop);
+    T acc = RedIdent<T>::get(rop_);
     for (uint64_t a = a0; a < a1; ++a)
-        acc = red_comb(redop, acc, mapf(mapop, base[row + a * inner]));
+        acc = red_comb(rop_, acc, mapf(mop_, base[row + a * inner]));
     partials[(o * nb + blockIdx.y) * inner + row] = acc;
 }
 
@@ -387,10 +398,10 @@ __global__ void reduce_dims_slices2(int redop, const T* __restrict__ partials,
     }
 }
 
-template <typename T>
-static int do_reduce_dims(int mapop, int redop, const T* src,
-                          uint64_t inner, uint64_t axis, uint64_t outer,
-                          T* dst, hipStream_t s) {
+template <typename T, int MOP, int ROP>
+static int do_reduce_dims_t(int mapop, int redop, const T* src,
+                            uint64_t inner, uint64_t axis, uint64_t outer,
+                            T* dst, hipStream_t s) {
     uint64_t total = inner * outer;
     if (total == 0) return 0;
     // axis == 0 still runs: the loop body never executes and dst gets
@@ -406,7 +417,8 @@ static int do_reduce_dims(int mapop, int redop, const T* src,
         if (rc) return rc;
         T* parts = (T*)st().partials;
         dim3 g((inner + RTPB - 1) / RTPB, nb, outer);
-        hipLaunchKernelGGL(reduce_dims_slices1<T>, g, dim3(RTPB), 0, s,
+        hipLaunchKernelGGL((reduce_dims_slices1<T, MOP, ROP>), g,
+                           dim3(RTPB), 0, s,
                            mapop, redop, src, inner, axis, nb, parts);
         DA_CHECK_HIP(hipGetLastError());
         uint64_t want = (total + RTPB - 1) / RTPB;
@@ -420,23 +432,51 @@ static int do_reduce_dims(int mapop, int redop, const T* src,
         // plenty of outputs, coalesced across inner: thread-per-output
         uint64_t want = (total + RTPB - 1) / RTPB;
         int g = (int)(want < 1 ? 1 : (want > 8192 ? 8192 : want));
-        hipLaunchKernelGGL(reduce_dims_threads<T>, dim3(g), dim3(RTPB), 0,
+        hipLaunchKernelGGL((reduce_dims_threads<T, MOP, ROP>), dim3(g),
+                           dim3(RTPB), 0,
                            s, mapop, redop, src, inner, axis, outer, dst);
     } else if (inner < 64 && axis >= 64 && total >= 16384) {
         // inner tiny (reduce over leading dim): wave-per-output, lanes
         // stride the axis (coalesced along the axis)
         uint64_t want = (total * 64 + RTPB - 1) / RTPB;
         int g = (int)(want < 1 ? 1 : (want > 8192 ? 8192 : want));
-        hipLaunchKernelGGL(reduce_dims_waves<T>, dim3(g), dim3(RTPB), 0,
+        hipLaunchKernelGGL((reduce_dims_waves<T, MOP, ROP>), dim3(g),
+                           dim3(RTPB), 0,
                            s, mapop, redop, src, inner, axis, outer, dst);
     } else {
         // few outputs / long axis: block-per-output
         int g = (int)(total < 1 ? 1 : (total > 8192 ? 8192 : total));
-        hipLaunchKernelGGL(reduce_dims_blocks<T>, dim3(g), dim3(RTPB), 0,
+        hipLaunchKernelGGL((reduce_dims_blocks<T, MOP, ROP>), dim3(g),
+                           dim3(RTPB), 0,
                            s, mapop, redop, src, inner, axis, outer, dst);
     }
     DA_CHECK_HIP(hipGetLastError());
     return 0;
+}
+
+// hot (mapop, redop) combos get constant-folded instantiations — the
+// dims-reduce analog of the map_fixed 2x recovery; everything else
+// rides the runtime-switch fallback.
+template <typename T>
+static int do_reduce_dims(int mapop, int redop, const T* src,
+                          uint64_t inner, uint64_t axis, uint64_t outer,
+                          T* dst, hipStream_t s) {
+    if (redop == DA_RED_ADD) {
+        if (mapop == DA_REDF_IDENTITY)
+            return do_reduce_dims_t<T, DA_REDF_IDENTITY, DA_RED_ADD>(
+                mapop, redop, src, inner, axis, outer, dst, s);
+        if (mapop == DA_REDF_ABS2)
+            return do_reduce_dims_t<T, DA_REDF_ABS2, DA_RED_ADD>(
+                mapop, redop, src, inner, axis, outer, dst, s);
+    }
+    if (redop == DA_RED_MAX && mapop == DA_REDF_IDENTITY)
+        return do_reduce_dims_t<T, DA_REDF_IDENTITY, DA_RED_MAX>(
+            mapop, redop, src, inner, axis, outer, dst, s);
+    if (redop == DA_RED_MIN && mapop == DA_REDF_IDENTITY)
+        return do_reduce_dims_t<T, DA_REDF_IDENTITY, DA_RED_MIN>(
+            mapop, redop, src, inner, axis, outer, dst, s);
+    return do_reduce_dims_t<T, -1, -1>(mapop, redop, src, inner, axis,
+                                       outer, dst, s);
 }
 
 int launch_reduce_dims(int mapop, int redop, const void* src,

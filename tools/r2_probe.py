@@ -119,6 +119,18 @@ def main():
             {"note": "16 B/elem algorithmic (R+W; mean row cached)"})
         X.close(); M.close(); D2.close()
 
+    if "dims" in legs:
+        m = 16384
+        X = dja.drand((m, m), "f64")
+        for red, name in [((0,), "dims0"), ((1,), "dims1")]:
+            def run(red=red):
+                R = dja.dsum_dims(X, red)
+                R.close()
+            t = bench(run, steps=8)
+            out("reduce_" + name, t * 1e3, m * m * 8 / t / 1e9,
+                {"note": "8 B/elem read, 16384^2 f64, whole-op"})
+        X.close()
+
     if "nt" in legs:
         # NT=0 baselines in this process
         D = dja.drand((n,), "f64")
