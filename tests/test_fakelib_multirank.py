@@ -339,6 +339,48 @@ def _scenario_spmd(rank, world, dja):
     sb.free(); rb.free()
 
 
+def _scenario_random_sweep(rank, world, dja):
+    """Randomized geometry sweep through the real orchestration code:
+    matmuls over random (m,k,n) x random grids (incl. degenerate cuts
+    from sz < chunks), dims-reductions over random axes, and random
+    getindex boxes — every rank derives the same random sequence from
+    the shared seed, so the collectives stay matched."""
+    rng = np.random.default_rng(2026)
+    grids = [(1, world), (world, 1)]
+    if world == 4:
+        grids += [(2, 2), (1, 2), (2, 1)]
+    for it in range(6):
+        m = int(rng.integers(1, 40))
+        kk = int(rng.integers(1, 30))
+        n = int(rng.integers(1, 40))
+        ga = _global_f64(m * kk, 1000 + it).reshape((m, kk), order="F")
+        gb = _global_f64(kk * n, 2000 + it).reshape((kk, n), order="F")
+        da_dist = grids[int(rng.integers(0, len(grids)))]
+        db_dist = grids[int(rng.integers(0, len(grids)))]
+        A = _slice_set(dja.DArray((m, kk), "f64", da_dist), ga)
+        B = _slice_set(dja.DArray((kk, n), "f64", db_dist), gb)
+        C = dja.dmatmul(A, B)
+        assert np.allclose(C.collect(), ga @ gb, rtol=1e-12, atol=1e-12), \
+            (it, (m, kk, n), da_dist, db_dist)
+        A.close(); B.close(); C.close()
+    for it in range(4):
+        nr = int(rng.integers(2, 30))
+        nc = int(rng.integers(2, 30))
+        g = _global_f64(nr * nc, 3000 + it).reshape((nr, nc), order="F")
+        dist = grids[int(rng.integers(0, len(grids)))]
+        D = _slice_set(dja.DArray((nr, nc), "f64", dist), g)
+        red = [(0,), (1,), (0, 1)][int(rng.integers(0, 3))]
+        R = dja.dsum_dims(D, red)
+        assert np.allclose(R.collect(), g.sum(axis=red, keepdims=True),
+                           rtol=1e-12), (it, (nr, nc), dist, red)
+        # random box fetch
+        r0 = int(rng.integers(0, nr)); r1 = int(rng.integers(r0 + 1, nr + 1))
+        c0 = int(rng.integers(0, nc)); c1 = int(rng.integers(c0 + 1, nc + 1))
+        got = dja.dgetindex(D, (r0, r1), (c0, c1))
+        assert np.array_equal(got, g[r0:r1, c0:c1])
+        D.close(); R.close()
+
+
 def _scenario_spmd_contexts(rank, world, dja):
     """Concurrent SPMD runs (test/spmd.jl:108-195 analog): 8 contexts
     whose ring-exchange operations INTERLEAVE round-robin — pairing
@@ -431,7 +473,7 @@ SCENARIOS_W2 = ["basic", "routing", "matmul_b_outside", "sort", "spmd",
 SCENARIOS_W4 = ["basic", "routing", "matmul", "matmul_nooverlap",
                 "matmul_emptyk", "matmul_nonidentity_raises",
                 "dims_reduce", "sort", "matvec", "halo", "spmd", "expr",
-                "spmd_contexts"]
+                "spmd_contexts", "random_sweep"]
 
 
 @pytest.mark.timeout(420)
