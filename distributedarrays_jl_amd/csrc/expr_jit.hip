@@ -175,6 +175,14 @@ int jit_get(const std::string& src, hipFunction_t* out) {
         *out = it->second.fn;
         return 0;
     }
+    if (g_jit_cache.size() >= 256) {
+        // bound module memory for workloads generating unbounded
+        // distinct programs; recompilation repopulates on demand.
+        // Drain the device first — a cached kernel may be in flight.
+        (void)hipDeviceSynchronize();
+        for (auto& kv : g_jit_cache) hipModuleUnload(kv.second.mod);
+        g_jit_cache.clear();
+    }
     const char* hdr_names[] = {"stdint.h", "darray_hip.h", "mapops.hpp",
                                "fastmath.hpp"};
     const char* stdint_stub =
