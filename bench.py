@@ -131,7 +131,9 @@ def main():
     # 1-D column distribution (2 GiB/GPU); other worlds weak-scale the
     # same bytes as a DVector.
     if world == 4:
-        shp, dist3 = (32768, 32768), (1, 4)
+        # side^2 == 4 * elems: 32768^2 at the default 2^28/GPU
+        side = 2 * int(round(n ** 0.5))
+        shp, dist3 = (side, side), (1, 4)
     else:
         shp, dist3 = (n * world,), (world,)
     A3 = dja.DArray(shp, "f64", dist3); A3.rand_()
@@ -182,7 +184,8 @@ def main():
     # ---------------- cfg5 (N==8): mapreduce(abs2,+,f32 2^31) ----------
     if world == 8:
         try:
-            F = dja.DArray((1 << 31,), "f32", (world,))
+            n5 = 8 * n     # 2^31 global at the default 2^28/GPU
+            F = dja.DArray((n5,), "f32", (world,))
             F.rand_()
             for _ in range(W):
                 dja.mapreduce("abs2", "add", F)
@@ -192,7 +195,7 @@ def main():
                 dja.mapreduce("abs2", "add", F)
             barrier()
             t5 = max_over_ranks(time.perf_counter() - t0)
-            extra["mapreduce_abs2_f32_gbs"] = (1 << 31) * 4.0 * K / t5 / 1e9
+            extra["mapreduce_abs2_f32_gbs"] = n5 * 4.0 * K / t5 / 1e9
             F.close()
             log("[bench] cfg5: %.1f GB/s"
                 % extra["mapreduce_abs2_f32_gbs"])
