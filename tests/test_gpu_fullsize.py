@@ -127,3 +127,27 @@ def test_fullsize_bcast_fma_property(dja):
     assert 0.25 * n < s < 0.75 * n
     for d in (A, B, D):
         d.close()
+
+
+def test_fullsize_expr_property(dja):
+    """Fused sin.(A) .+ B .* c at 2 GiB (the hipRTC JIT path at bench
+    size): exact spot checks vs libm-at-tolerance + a magnitude bound
+    on the whole-array sum."""
+    from distributedarrays_jl_amd import expr as E
+    n = 1 << 28
+    A = dja.drand((n,), "f64")
+    B = dja.drand((n,), "f64", seed_base=4321)
+    D = dja.dzeros((n,))
+    E.materialize_(D, E.sin(E.ref(A)) + E.ref(B) * 0.5)
+    rng = np.random.default_rng(3)
+    idxs = sorted(int(i) for i in rng.integers(0, n, 128))
+    got = _read_elems(D, idxs)
+    a = _philox_at(1234, np.array(idxs))
+    b = _philox_at(4321, np.array(idxs))
+    ref = np.sin(a) + b * 0.5
+    assert np.allclose(got, ref, rtol=5e-16, atol=5e-16)
+    # E[sin(U)] + E[U]/2 = (1-cos(1)) + 0.25 ~= 0.7097
+    s = dja.dsum(D) / n
+    assert abs(s - ((1 - np.cos(1.0)) + 0.25)) < 1e-3
+    for d in (A, B, D):
+        d.close()
