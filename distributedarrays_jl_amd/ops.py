@@ -251,14 +251,26 @@ def ddot(x, y):
 
 
 def dnorm(x, p=2):
-    """norm — linalg.jl:47-52 (p in {1, 2, inf})."""
+    """norm(x, p) for any real p — linalg.jl:47-52 (the reference runs
+    Base.norm on each localpart then norms the partials; here the
+    closed forms: p-power sum via one fused |x|^p kernel)."""
     if p == 2:
         return float(np.sqrt(mapreduce("abs2", "add", x)))
     if p == 1:
         return mapreduce("abs", "add", x)
     if p == float("inf"):
         return mapreduce("abs", "max", x)
-    raise DArrayError("norm: p=%r not supported" % p)
+    if p == float("-inf"):
+        return mapreduce("abs", "min", x)
+    if p == 0:
+        return float(dcount("nonzero", x))   # Julia norm(x, 0)
+    from . import expr as E
+    tmp = x.similar()
+    E.materialize_(tmp, E.abs(E.ref(x)) ** float(p))
+    try:
+        return float(mapreduce("identity", "add", tmp)) ** (1.0 / p)
+    finally:
+        tmp.close()
 
 
 # --------------------------------------------------------------- matmul
@@ -655,22 +667,24 @@ def dmatvec(A, x, alpha=1.0):
     (linalg.jl:91: xj shipped per tile), partial vectors travel to the
     y owner (rank i = procs(A)[i,1]) and accumulate ascending j."""
     import numpy as np
-    if A.dtype != "f64" or A.ndims != 2:
-        raise DArrayError("dmatvec: 2-D f64 only")
+    if A.dtype not in ("f64", "f32") or A.ndims != 2:
+        raise DArrayError("dmatvec: 2-D f64/f32 only")
     if A.ranks != list(range(A.nchunks)):
         raise DArrayError(
             "dmatvec: A must have identity chunk->rank mapping "
             "(copy() a reduction result first)")
-    x = np.ascontiguousarray(np.asarray(x, dtype=np.float64))
+    gemm_fn = lib.da_gemm_f64 if A.dtype == "f64" else lib.da_gemm_f32
+    x = np.ascontiguousarray(np.asarray(x,
+                             dtype=NUMPY_DTYPES[A.dtype]))
     if x.shape != (A.dims[1],):
         raise DArrayError("dmatvec: x length %d != %d"
                           % (x.shape[0], A.dims[1]))
     m = A.dims[0]
     I, J = A.dist
-    y = DArray((m,), "f64", (I,))
+    y = DArray((m,), A.dtype, (I,))
     y.fill_(0.0)
     r = A.rank
-    esz = 8
+    esz = DTYPE_SIZE[A.dtype]
     partial = None
     if A.lchunk is not None and A.lnumel:
         i, j = r % I, r // I
@@ -681,8 +695,8 @@ def dmatvec(A, x, alpha=1.0):
                          xj.size * esz))
         mloc, kloc = A.lshape
         partial = _Buf(mloc * esz)
-        check(lib.da_gemm_f64(partial.p, A._ptr(), xbuf.p,
-                              mloc, 1, kloc, mloc, kloc, mloc, 1.0, 0.0))
+        check(gemm_fn(partial.p, A._ptr(), xbuf.p,
+                      mloc, 1, kloc, mloc, kloc, mloc, 1.0, 0.0))
         xbuf.free()
 
     sends, recvs = [], {}
@@ -706,7 +720,7 @@ def dmatvec(A, x, alpha=1.0):
             src = r + I * j
             buf = partial if src == r else recvs[src]
             check(lib.da_add(y._ptr(), buf.p, float(alpha), y.lnumel,
-                             DTYPES["f64"]))
+                             DTYPES[A.dtype]))
     check(lib.da_synchronize())
     if partial is not None:
         partial.free()
@@ -726,20 +740,22 @@ def dmatvec_adj(A, x, alpha=1.0):
     The local A_loc'*xj is one da_gemm_f64 with the vector as the 1-row
     left operand: C(1 x kloc) = x_row(1 x mloc) * A_loc(mloc x kloc)."""
     import numpy as np
-    if A.dtype != "f64" or A.ndims != 2:
-        raise DArrayError("dmatvec_adj: 2-D f64 only")
+    if A.dtype not in ("f64", "f32") or A.ndims != 2:
+        raise DArrayError("dmatvec_adj: 2-D f64/f32 only")
     if A.ranks != list(range(A.nchunks)):
         raise DArrayError(
             "dmatvec_adj: A must have identity chunk->rank mapping")
-    x = np.ascontiguousarray(np.asarray(x, dtype=np.float64))
+    gemm_fn = lib.da_gemm_f64 if A.dtype == "f64" else lib.da_gemm_f32
+    x = np.ascontiguousarray(np.asarray(x,
+                             dtype=NUMPY_DTYPES[A.dtype]))
     if x.shape != (A.dims[0],):
         raise DArrayError("dmatvec_adj: x length %d != %d"
                           % (x.shape[0], A.dims[0]))
     I, J = A.dist
-    y = DArray((A.dims[1],), "f64", (J,))
+    y = DArray((A.dims[1],), A.dtype, (J,))
     y.fill_(0.0)
     r = A.rank
-    esz = 8
+    esz = DTYPE_SIZE[A.dtype]
     partial = None
     kloc = 0
     if A.lchunk is not None and A.lnumel:
@@ -750,8 +766,8 @@ def dmatvec_adj(A, x, alpha=1.0):
                          xi.size * esz))
         mloc, kloc = A.lshape
         partial = _Buf(max(kloc, 1) * esz)
-        check(lib.da_gemm_f64(partial.p, xbuf.p, A._ptr(),
-                              1, kloc, mloc, 1, mloc, 1, 1.0, 0.0))
+        check(gemm_fn(partial.p, xbuf.p, A._ptr(),
+                      1, kloc, mloc, 1, mloc, 1, 1.0, 0.0))
         xbuf.free()
 
     sends, recvs = [], {}
@@ -776,7 +792,7 @@ def dmatvec_adj(A, x, alpha=1.0):
             src = i + I * r
             buf = partial if src == r else recvs[src]
             check(lib.da_add(y._ptr(), buf.p, float(alpha), y.lnumel,
-                             DTYPES["f64"]))
+                             DTYPES[A.dtype]))
     check(lib.da_synchronize())
     if partial is not None:
         partial.free()

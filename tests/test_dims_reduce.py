@@ -159,3 +159,27 @@ def test_ops_on_dims_reduction_result():
     assert np.allclose(out.collect(), ref, rtol=1e-12)
     assert abs(dja.dsum(R) - x.sum()) < 1e-9
     out.close(); R.close(); d.close()
+
+
+@pytest.mark.gpu
+def test_gpu_matvec_f32_and_norm_p():
+    import distributedarrays_jl_amd as dja
+    dja.comm.init()
+    m, k = 96, 64
+    A = np.asfortranarray(philox.fill_uniform_f32(m * k, 21)
+                          .reshape(m, k, order="F"))
+    x = philox.fill_uniform_f32(k, 22)
+    dA = dja.distribute(A)
+    y = dja.dmatvec(dA, x)
+    assert np.allclose(y.collect(), A @ x, rtol=1e-5)
+    xr = philox.fill_uniform_f32(m, 23)
+    ya = dja.dmatvec_adj(dA, xr)
+    assert np.allclose(ya.collect(), A.T @ xr, rtol=1e-5)
+    y.close(); ya.close(); dA.close()
+    # general-p norm (linalg.jl:47-52)
+    v = philox.fill_uniform_f64(5001, 24) - 0.5
+    dv = dja.distribute(v)
+    ref = float(np.sum(np.abs(v) ** 3.5) ** (1 / 3.5))
+    assert abs(dja.dnorm(dv, 3.5) - ref) <= 1e-12 * ref
+    assert dja.dnorm(dv, 0) == float(np.count_nonzero(v))
+    dv.close()

@@ -286,6 +286,27 @@ def _scenario_matvec(rank, world, dja):
     y2 = dja.dmatvec_adj(A, x2, alpha=0.5)
     assert np.allclose(y2.collect(), 0.5 * (ga.T @ x2), rtol=1e-12)
     A.close(); y.close(); y2.close()
+    # f32 matvec + adjoint
+    ga32 = ga.astype(np.float32)
+    A32 = dja.DArray((m, kk), "f32", (2, world // 2))
+    _slice_set(A32, ga32)
+    y32 = dja.dmatvec(A32, x.astype(np.float32))
+    assert np.allclose(y32.collect(), ga32 @ x.astype(np.float32),
+                       rtol=1e-5)
+    y32a = dja.dmatvec_adj(A32, x2.astype(np.float32))
+    assert np.allclose(y32a.collect(), ga32.T @ x2.astype(np.float32),
+                       rtol=1e-5)
+    A32.close(); y32.close(); y32a.close()
+    # general-p norm (linalg.jl:47-52): sum(|x|^p)^(1/p)
+    gv = _global_f64(40, 21) - 0.5
+    V = _slice_set(dja.DArray((40,), "f64"), gv)
+    got = dja.dnorm(V, 3.0)
+    ref = float(np.sum(np.abs(gv) ** 3.0) ** (1.0 / 3.0))
+    assert abs(got - ref) <= 1e-12 * ref
+    assert dja.dnorm(V, 0) == float(np.count_nonzero(gv))
+    assert abs(dja.dnorm(V, float("-inf"))
+               - np.min(np.abs(gv))) < 1e-15
+    V.close()
 
 
 def _scenario_halo(rank, world, dja):
