@@ -165,6 +165,28 @@ def scale_(a, s):
     return a
 
 
+def map_localparts(f, *ds):
+    """map_localparts(f, d...) — mapreduce.jl:137-169: apply an
+    ARBITRARY host function to whole localparts (numpy in, numpy out;
+    the reference runs arbitrary Julia closures on its CPU localparts
+    the same way).  This is a host-boundary escape hatch: the chunk
+    round-trips HBM -> host -> HBM.  Compositions of the op tables
+    should use expr.materialize_ instead (fused, device-resident)."""
+    res = f(*[d.localpart() for d in ds])
+    out = ds[0].similar()
+    out.set_localpart(np.asfortranarray(
+        np.asarray(res, dtype=np.dtype(NUMPY_DTYPES[out.dtype]))))
+    return out
+
+
+def map_localparts_(f, d):
+    """map_localparts!(f, d) — in-place variant (mapreduce.jl:156-162)."""
+    d.set_localpart(np.asfortranarray(
+        np.asarray(f(d.localpart()),
+                   dtype=np.dtype(NUMPY_DTYPES[d.dtype]))))
+    return d
+
+
 # ------------------------------------------------------------ reductions
 _CTYPE = {"f64": ctypes.c_double, "f32": ctypes.c_float,
           "i64": ctypes.c_int64}

@@ -478,6 +478,26 @@ def _scenario_expr(rank, world, dja):
         d.close()
 
 
+def _scenario_map_localparts(rank, world, dja):
+    """mapreduce.jl:137-169: arbitrary host function per localpart."""
+    n = 200
+    g = _global_f64(n, 30)
+    D = _slice_set(dja.DArray((n,), "f64"), g)
+    R = dja.map_localparts(lambda lp: np.cumsum(lp), D)
+    ref_chunks = []
+    for (lo, hi), in D.idxs:
+        ref_chunks.append(np.cumsum(g[lo:hi]))
+    assert np.allclose(R.collect(), np.concatenate(ref_chunks), rtol=0)
+    dja.map_localparts_(lambda lp: lp * 2.0, D)
+    assert np.allclose(D.collect(), 2.0 * g, rtol=0)
+    # binary form
+    E2 = _slice_set(dja.DArray((n,), "f64"), g)
+    S = dja.map_localparts(lambda a, b: a + b, D, E2)
+    assert np.allclose(S.collect(), 3.0 * g, rtol=0)
+    for d in (D, R, E2, S):
+        d.close()
+
+
 def _scenario_scalar_index(rank, world, dja):
     n = 40
     g = _global_f64(n, 17)
@@ -494,7 +514,7 @@ SCENARIOS_W2 = ["basic", "routing", "matmul_b_outside", "sort", "spmd",
 SCENARIOS_W4 = ["basic", "routing", "matmul", "matmul_nooverlap",
                 "matmul_emptyk", "matmul_nonidentity_raises",
                 "dims_reduce", "sort", "matvec", "halo", "spmd", "expr",
-                "spmd_contexts", "random_sweep"]
+                "spmd_contexts", "random_sweep", "map_localparts"]
 
 
 @pytest.mark.timeout(420)
