@@ -25,7 +25,8 @@ from .ops import (map_, dmap, map2_, elementwise, map2_scalar_, elementwise_scal
                   dextrema, dmean, dcount, dall, dany, ddot, dnorm, dmatmul, dreduce_dims,
                   dsum_dims, dprod_dims, dmaximum_dims, dminimum_dims,
                   dmean_dims, dmatvec, dmatvec_adj, gather_box, map_general,
-                  map_localparts, map_localparts_,
+                  map_localparts, map_localparts_, redistribute,
+                  dmapslices, dppeval,
                   map2_general,
                   broadcast_fma_general, dsort, dtranspose, ddiag_lmul, ddiag_rmul,
                   dgetindex, dmul_)
@@ -40,7 +41,8 @@ __all__ = [
     "dminimum", "dextrema", "dmean", "dcount", "dall", "dany", "ddot", "dnorm", "dmatmul",
     "dreduce_dims", "dsum_dims", "dprod_dims", "dmaximum_dims",
     "dminimum_dims", "dmean_dims", "dmatvec", "dmatvec_adj",
-    "map_localparts", "map_localparts_",
+    "map_localparts", "map_localparts_", "redistribute",
+    "dmapslices", "dppeval",
     "map2_general",
     "gather_box", "map_general", "broadcast_fma_general", "dsort",
     "dtranspose", "ddiag_lmul", "ddiag_rmul", "dgetindex", "dmul_",

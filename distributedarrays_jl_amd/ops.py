@@ -937,6 +937,167 @@ def _dest_boxes(dest):
     return boxes
 
 
+def redistribute(D, dist):
+    """A new DArray with the same global content on a different chunk
+    grid — the reference's re-distribution constructor pattern
+    (mapslices builds one at mapreduce.jl:196-202 via the init
+    constructor pulling D[I...]).  Device-resident: each new-chunk
+    owner gathers its box over xGMI (gather_box); collective."""
+    dist = tuple(int(c) for c in dist)
+    out = DArray(D.dims, D.dtype, dist)
+    boxes = [None] * D.nranks
+    for c, r in enumerate(out.ranks):
+        boxes[r] = out.idxs[c]
+    buf, shape = gather_box(D, boxes)
+    if buf is not None and out.lnumel:
+        check(lib.da_d2d(out._ptr(), buf.p,
+                         out.lnumel * DTYPE_SIZE[out.dtype]))
+        check(lib.da_synchronize())
+    if buf is not None:
+        buf.free()
+    return out
+
+
+def dmapslices(f, D, dims):
+    """mapslices(f, D; dims) — mapreduce.jl:191-208: if any sliced dim
+    is distributed, redistribute so slices are whole per chunk (the
+    reference builds a DD with p[dims]=1), then apply numpy's
+    apply-over-slices on each localpart (host boundary, like
+    map_localparts) and assemble the result DArray on the non-sliced
+    chunk grid."""
+    import numpy as np
+    if isinstance(dims, int):
+        dims = (dims,)
+    dims = tuple(sorted(set(int(a) for a in dims)))
+    nd = D.ndims
+    if any(a < 0 or a >= nd for a in dims):
+        raise DArrayError("dmapslices: bad dims %r" % (dims,))
+    if any(D.dist[a] != 1 for a in dims):
+        nondims = [a for a in range(nd) if a not in dims]
+        p = [1] * nd
+        sub = geometry.defaultdist([D.dims[a] for a in nondims],
+                                   D.nranks)
+        for a, c in zip(nondims, sub):
+            p[a] = c
+        DD = redistribute(D, p)
+        try:
+            return dmapslices(f, DD, dims)
+        finally:
+            DD.close()
+
+    lp = D.localpart()
+
+    def apply_local(arr):
+        # evaluate f on every dims-slice of arr, stack the results in
+        # place of the sliced dims (Julia mapslices semantics for
+        # same-shape outputs; general reshaping follows Base)
+        nondims = [a for a in range(nd) if a not in dims]
+        it_shape = [arr.shape[a] for a in nondims]
+        probe_idx = [slice(None)] * nd
+        for a in nondims:
+            probe_idx[a] = 0
+        r1 = np.asarray(f(np.asarray(arr[tuple(probe_idx)])))
+        out_shape = list(arr.shape)
+        rs = list(r1.shape) + [1] * (len(dims) - r1.ndim)
+        for a, s in zip(dims, rs):
+            out_shape[a] = s
+        out = np.empty(out_shape, dtype=r1.dtype, order="F")
+        for pos in np.ndindex(*it_shape):
+            idx = [slice(None)] * nd
+            for a, v in zip(nondims, pos):
+                idx[a] = v
+            out[tuple(idx)] = np.asarray(
+                f(np.asarray(arr[tuple(idx)]))).reshape(
+                [out_shape[a] for a in dims], order="A")
+        return out
+
+    local = (apply_local(lp) if D.lnumel else
+             np.empty([0] * nd, order="F"))
+    # result dims: sliced dims take f's output size (agreed across
+    # ranks via the shared metadata — f must be shape-uniform, as in
+    # the reference), non-sliced dims keep D's
+    if D.lnumel:
+        out_sizes = list(local.shape)
+    else:
+        out_sizes = [0] * nd
+    import torch.distributed as td
+    if D.nranks > 1 and td.is_initialized():
+        lst = [None] * D.nranks
+        td.all_gather_object(lst, out_sizes if D.lnumel else None)
+        out_sizes = next(s for s in lst if s is not None)
+    rdims = tuple(out_sizes[a] if a in dims else D.dims[a]
+                  for a in range(nd))
+    R = DArray(rdims, D.dtype, D.dist, ranks=list(D.ranks))
+    if R.lnumel:
+        R.set_localpart(np.asfortranarray(
+            np.asarray(local, dtype=np.dtype(NUMPY_DTYPES[D.dtype]))))
+    return R
+
+
+def dppeval(f, *Ds, dim=None):
+    """ppeval(f, D...; dim) — mapreduce.jl:258-323: evaluate f on the
+    dim-slices of each argument (default: last dim for DArrays, whole
+    array broadcast for numpy arguments), stacking results along a new
+    last dimension distributed like the first DArray's sliced dim.
+    Host boundary (arbitrary f), like the reference's per-worker
+    _ppeval on localparts."""
+    import numpy as np
+    if not Ds or not isinstance(Ds[0], DArray):
+        raise DArrayError("dppeval: first argument must be a DArray "
+                          "(procs(D[1]) in the reference)")
+    dax = []
+    for a in Ds:
+        dax.append(a.ndims - 1 if isinstance(a, DArray) else 0)
+    if dim is not None:
+        dax = [d - 1 if d > 0 else 0 for d in dim]   # 1-based like ref
+    first = Ds[0]
+    for a, d in zip(Ds, dax):
+        if isinstance(a, DArray):
+            for ax in range(a.ndims):
+                if ax != d and a.dist[ax] != 1:
+                    raise DArrayError(
+                        "dppeval: dimension %d is distributed; must be "
+                        "whole per chunk" % ax)
+    locals_ = [a.localpart() if isinstance(a, DArray) else np.asarray(a)
+               for a in Ds]
+
+    def slice_at(arr, d, i, is_d):
+        if not is_d:
+            return arr
+        idx = [slice(None)] * arr.ndim
+        idx[d] = i
+        return arr[tuple(idx)]
+
+    n_loc = locals_[0].shape[dax[0]]
+    outs = []
+    for i in range(n_loc):
+        args = [slice_at(lp, d, i, isinstance(a, DArray))
+                for lp, d, a in zip(locals_, dax, Ds)]
+        outs.append(np.asarray(f(*args)))
+    if outs:
+        local = np.stack(outs, axis=-1)
+    else:
+        local = np.empty((0,), order="F")
+    # result: (fshape..., global dimlen), distributed along the last
+    # dim with the first DArray's cuts on its sliced dim
+    fshape = list(outs[0].shape) if outs else []
+    import torch.distributed as td
+    if first.nranks > 1 and td.is_initialized():
+        lst = [None] * first.nranks
+        td.all_gather_object(lst, fshape if outs else None)
+        fshape = next(s for s in lst if s is not None)
+    gdim = first.dims[dax[0]]
+    nchunks = first.dist[dax[0]]
+    rdims = tuple(fshape) + (gdim,)
+    rdist = tuple([1] * len(fshape)) + (nchunks,)
+    R = DArray(rdims, first.dtype, rdist, ranks=list(first.ranks))
+    if R.lnumel:
+        R.set_localpart(np.asfortranarray(
+            np.asarray(local, dtype=np.dtype(NUMPY_DTYPES[R.dtype]))
+            .reshape(R.lshape, order="A")))
+    return R
+
+
 def dgetindex(A, *ranges):
     """D[I...] for range indexing -> numpy array on every rank
     (Array(view(A, I...)), the makelocal contract darray.jl:346-368;

@@ -183,3 +183,32 @@ def test_gpu_matvec_f32_and_norm_p():
     assert abs(dja.dnorm(dv, 3.5) - ref) <= 1e-12 * ref
     assert dja.dnorm(dv, 0) == float(np.count_nonzero(v))
     dv.close()
+
+
+@pytest.mark.gpu
+def test_gpu_mapslices_ppeval_redistribute():
+    """mapreduce.jl:191-323 host-boundary veneers on the GPU path."""
+    import distributedarrays_jl_amd as dja
+    dja.comm.init()
+    g = np.asfortranarray(philox.fill_uniform_f64(24 * 10, 31)
+                          .reshape(24, 10, order="F"))
+    D = dja.distribute(g)
+    R = dja.redistribute(D, (1, 1))
+    assert np.array_equal(R.collect(), g)
+    R.close()
+    M = dja.dmapslices(lambda col: np.cumsum(col), D, (0,))
+    assert np.allclose(M.collect(), np.cumsum(g, axis=0), rtol=0)
+    M.close()
+    M2 = dja.dmapslices(lambda col: np.array([col.sum()]), D, (0,))
+    assert np.allclose(M2.collect(), g.sum(axis=0, keepdims=True),
+                       rtol=1e-12)
+    M2.close(); D.close()
+    g3 = np.asfortranarray(philox.fill_uniform_f64(4 * 4 * 6, 32)
+                           .reshape(4, 4, 6, order="F"))
+    A3 = dja.distribute(g3)
+    w = np.asfortranarray(philox.fill_uniform_f64(16, 33)
+                          .reshape(4, 4, order="F"))
+    P = dja.dppeval(lambda s, b: s @ b, A3, w)
+    ref = np.stack([g3[:, :, i] @ w for i in range(6)], axis=-1)
+    assert np.allclose(P.collect(), ref, rtol=1e-12)
+    P.close(); A3.close()

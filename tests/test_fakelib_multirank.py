@@ -498,6 +498,52 @@ def _scenario_map_localparts(rank, world, dja):
         d.close()
 
 
+def _scenario_slices(rank, world, dja):
+    """redistribute + mapslices + ppeval (mapreduce.jl:191-323)."""
+    nr, nc = 12, 4 * world
+    g = _global_f64(nr * nc, 31).reshape((nr, nc), order="F")
+    D = _slice_set(dja.DArray((nr, nc), "f64", (1, world)), g)
+    # redistribute to a row split and back
+    R = dja.redistribute(D, (world, 1))
+    assert np.array_equal(R.collect(), g)
+    R2 = dja.redistribute(R, (1, world))
+    assert np.array_equal(R2.collect(), g)
+    R.close(); R2.close()
+    # mapslices over the undistributed dim (column-wise f)
+    M = dja.dmapslices(lambda col: np.cumsum(col), D, (0,))
+    assert np.allclose(M.collect(), np.cumsum(g, axis=0), rtol=0)
+    M.close()
+    # mapslices over the DISTRIBUTED dim -> internal redistribute
+    M2 = dja.dmapslices(lambda row: row - row.mean(), D, (1,))
+    ref = g - g.mean(axis=1, keepdims=True)
+    assert np.allclose(M2.collect(), ref, rtol=1e-12)
+    M2.close()
+    # shape-changing f: reduce each column to a scalar
+    M3 = dja.dmapslices(lambda col: np.array([col.sum()]), D, (0,))
+    assert M3.dims == (1, nc)
+    assert np.allclose(M3.collect(), g.sum(axis=0, keepdims=True),
+                       rtol=1e-12)
+    M3.close()
+    # ppeval: slices along the last dim; second arg broadcast
+    A3 = dja.DArray((3, 3, 2 * world), "f64", (1, 1, world))
+    g3 = _global_f64(9 * 2 * world, 32).reshape((3, 3, 2 * world),
+                                                order="F")
+    _slice_set(A3, g3)
+    w = _global_f64(9, 33).reshape((3, 3), order="F")
+    P = dja.dppeval(lambda s, b: s @ b, A3, w)
+    assert P.dims == (3, 3, 2 * world)
+    ref = np.stack([g3[:, :, i] @ w for i in range(2 * world)], axis=-1)
+    assert np.allclose(P.collect(), ref, rtol=1e-12)
+    P.close()
+    # ppeval scalar-result f: per-slice trace
+    T = dja.dppeval(lambda s: np.trace(s), A3)
+    assert np.allclose(
+        T.collect().ravel(),
+        np.array([np.trace(g3[:, :, i]) for i in range(2 * world)]),
+        rtol=0)
+    T.close(); A3.close(); D.close()
+
+
 def _scenario_scalar_index(rank, world, dja):
     n = 40
     g = _global_f64(n, 17)
@@ -514,7 +560,8 @@ SCENARIOS_W2 = ["basic", "routing", "matmul_b_outside", "sort", "spmd",
 SCENARIOS_W4 = ["basic", "routing", "matmul", "matmul_nooverlap",
                 "matmul_emptyk", "matmul_nonidentity_raises",
                 "dims_reduce", "sort", "matvec", "halo", "spmd", "expr",
-                "spmd_contexts", "random_sweep", "map_localparts"]
+                "spmd_contexts", "random_sweep", "map_localparts",
+                "slices"]
 
 
 @pytest.mark.timeout(420)
