@@ -360,13 +360,13 @@ def _scenario_spmd(rank, world, dja):
     sb.free(); rb.free()
 
 
-def _scenario_random_sweep(rank, world, dja):
+def _scenario_random_sweep(rank, world, dja, seed=2026):
     """Randomized geometry sweep through the real orchestration code:
     matmuls over random (m,k,n) x random grids (incl. degenerate cuts
     from sz < chunks), dims-reductions over random axes, and random
     getindex boxes — every rank derives the same random sequence from
     the shared seed, so the collectives stay matched."""
-    rng = np.random.default_rng(2026)
+    rng = np.random.default_rng(seed)
     grids = [(1, world), (world, 1)]
     if world == 4:
         grids += [(2, 2), (1, 2), (2, 1)]
@@ -573,4 +573,18 @@ def test_world2(tmp_path, name):
 @pytest.mark.timeout(420)
 @pytest.mark.parametrize("name", SCENARIOS_W4)
 def test_world4(tmp_path, name):
+    _spawn(tmp_path, 4, name)
+
+
+def _scenario_random_sweep_b(rank, world, dja):
+    _scenario_random_sweep(rank, world, dja, seed=31337)
+
+
+def _scenario_random_sweep_c(rank, world, dja):
+    _scenario_random_sweep(rank, world, dja, seed=777)
+
+
+@pytest.mark.timeout(420)
+@pytest.mark.parametrize("name", ["random_sweep_b", "random_sweep_c"])
+def test_world4_sweep_seeds(tmp_path, name):
     _spawn(tmp_path, 4, name)
