@@ -61,6 +61,7 @@ _sigs = {
     "da_axpby": ([ptr, ptr, f64, f64, u64, i32], i32),
     "da_add": ([ptr, ptr, f64, u64, i32], i32),
     "da_scale": ([ptr, f64, u64, i32], i32),
+    "da_cast": ([ptr, i32, ptr, i32, u64], i32),
     "da_reduce": ([i32, i32, ptr, u64, i32, ptr], i32),
     "da_reduce_dims": ([i32, i32, ptr, u64, u64, u64, i32, ptr], i32),
     "da_transpose": ([ptr, ptr, u64, u64, i32], i32),

@@ -335,6 +335,12 @@ int da_scale(void* a, double s, uint64_t n, int dtype) {
     return launch_scale(a, s, n, dtype, st().stream);
 }
 
+int da_cast(void* dst, int dst_dtype, const void* src, int src_dtype,
+            uint64_t n) {
+    DA_REQUIRE_INIT();
+    return launch_cast(dst, dst_dtype, src, src_dtype, n, st().stream);
+}
+
 int da_reduce(int mapop, int redop, const void* src, uint64_t n, int dtype,
               void* out) {
     DA_REQUIRE_INIT();

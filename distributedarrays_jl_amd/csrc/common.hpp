@@ -119,6 +119,8 @@ int launch_axpby(void* y, const void* x, double alpha, double beta,
 int launch_add(void* dest, const void* src, double scale, uint64_t n,
                int dtype, hipStream_t s);
 int launch_scale(void* a, double s_, uint64_t n, int dtype, hipStream_t s);
+int launch_cast(void* dst, int dst_dtype, const void* src, int src_dtype,
+                uint64_t n, hipStream_t s);
 int launch_reduce(int mapop, int redop, const void* src, uint64_t n,
                   int dtype, void* out_host, hipStream_t s);
 int launch_reduce_dims(int mapop, int redop, const void* src,

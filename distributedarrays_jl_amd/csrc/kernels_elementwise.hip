@@ -620,4 +620,61 @@ int launch_scale(void* a, double s_, uint64_t n, int dtype, hipStream_t s) {
     return 0;
 }
 
+
+// ------------------------------------------------------------------ cast
+// dtype conversion (the DArray{T2}(D) / convert family): dst[i] =
+// (TD)src[i].  float->i64 rounds half-even (Julia round(Int, x)); the
+// strict InexactError convert is a host-side concern.
+template <typename TD, typename TS>
+__global__ void cast_kernel(TD* __restrict__ dst,
+                            const TS* __restrict__ src, uint64_t n) {
+    uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t j = i; j < n; j += stride) dst[j] = (TD)src[j];
+}
+
+template <>
+__global__ void cast_kernel<int64_t, double>(int64_t* __restrict__ dst,
+                                             const double* __restrict__ src,
+                                             uint64_t n) {
+    uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t j = i; j < n; j += stride)
+        dst[j] = (int64_t)rint(src[j]);
+}
+
+template <>
+__global__ void cast_kernel<int64_t, float>(int64_t* __restrict__ dst,
+                                            const float* __restrict__ src,
+                                            uint64_t n) {
+    uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t j = i; j < n; j += stride)
+        dst[j] = (int64_t)rintf(src[j]);
+}
+
+int launch_cast(void* dst, int dst_dtype, const void* src, int src_dtype,
+                uint64_t n, hipStream_t s) {
+    if (n == 0) return 0;
+    if (dst_dtype == src_dtype)
+        return set_err(-3, "da_cast: same dtype (use da_d2d)");
+    int g = nblocks(n);
+#define DA_CAST(TD, TS) \
+    hipLaunchKernelGGL((cast_kernel<TD, TS>), dim3(g), dim3(TPB), 0, s, \
+                       (TD*)dst, (const TS*)src, n)
+    switch (dst_dtype * 4 + src_dtype) {
+    case DA_F64 * 4 + DA_F32: DA_CAST(double, float); break;
+    case DA_F64 * 4 + DA_I64: DA_CAST(double, int64_t); break;
+    case DA_F32 * 4 + DA_F64: DA_CAST(float, double); break;
+    case DA_F32 * 4 + DA_I64: DA_CAST(float, int64_t); break;
+    case DA_I64 * 4 + DA_F64: DA_CAST(int64_t, double); break;
+    case DA_I64 * 4 + DA_F32: DA_CAST(int64_t, float); break;
+    default: return set_err(-3, "da_cast: bad dtypes %d<-%d",
+                            dst_dtype, src_dtype);
+    }
+#undef DA_CAST
+    DA_CHECK_HIP(hipGetLastError());
+    return 0;
+}
+
 } // namespace da

@@ -165,6 +165,21 @@ def scale_(a, s):
     return a
 
 
+def dcast(d, dtype):
+    """DArray{T2}(D) — elementwise dtype conversion on device.
+    float -> i64 rounds half-even (round(Int, x)); the strict
+    InexactError convert is a host-side concern."""
+    if dtype == d.dtype:
+        return d.copy()
+    if dtype not in DTYPES:
+        raise DArrayError("dcast: bad dtype %r" % dtype)
+    out = d.similar(dtype)
+    if out.lnumel:
+        check(lib.da_cast(out._ptr(), DTYPES[dtype], d._ptr(),
+                          DTYPES[d.dtype], out.lnumel))
+    return out
+
+
 def map_localparts(f, *ds):
     """map_localparts(f, d...) — mapreduce.jl:137-169: apply an
     ARBITRARY host function to whole localparts (numpy in, numpy out;

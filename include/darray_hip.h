@@ -148,6 +148,9 @@ int da_axpby(void* y, const void* x, double alpha, double beta,
 int da_add(void* dest, const void* src, double scale,
            uint64_t n, int dtype);                           /* add!, linalg.jl:62-76 */
 int da_scale(void* a, double s, uint64_t n, int dtype);      /* rmul!, linalg.jl:54-59 */
+/* dtype conversion (DArray{T2}(D) family); float->i64 rounds half-even */
+int da_cast(void* dst, int dst_dtype, const void* src, int src_dtype,
+            uint64_t n);
 
 /* ---- reductions -------------------------------------------------------*/
 /* Local (per-chunk) mapreduce stage: LDS + wavefront-shuffle tree.

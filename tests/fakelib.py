@@ -210,6 +210,15 @@ class FakeLib:
         _tv(dst, n, dt)[:] = np.asfortranarray(out).ravel(order="F")
         return 0
 
+    def da_cast(self, dst, dst_dtype, src, src_dtype, n):
+        dd, sd = _NPDT[int(dst_dtype)], _NPDT[int(src_dtype)]
+        sv = _tv(src, n, sd)
+        if dd.kind == "i" and sd.kind == "f":
+            _tv(dst, n, dd)[:] = np.rint(sv).astype(dd)
+        else:
+            _tv(dst, n, dd)[:] = sv.astype(dd)
+        return 0
+
     def da_expr_jit_state(self):
         return 0      # no JIT in the fake (numpy evaluator only)
 

@@ -588,3 +588,27 @@ def _scenario_random_sweep_c(rank, world, dja):
 @pytest.mark.parametrize("name", ["random_sweep_b", "random_sweep_c"])
 def test_world4_sweep_seeds(tmp_path, name):
     _spawn(tmp_path, 4, name)
+
+
+def _scenario_cast(rank, world, dja):
+    """DArray{T2}(D) conversion chain: f64 -> f32 -> f64, f64 -> i64
+    (half-even) -> sum, then aligned ops on the result."""
+    n = 500
+    g = _global_f64(n, 41) * 100.0
+    D = _slice_set(dja.DArray((n,), "f64"), g)
+    F = dja.dcast(D, "f32")
+    assert np.array_equal(F.collect(), g.astype(np.float32))
+    B = dja.dcast(F, "f64")
+    assert np.array_equal(B.collect(), g.astype(np.float32)
+                          .astype(np.float64))
+    I = dja.dcast(D, "i64")
+    assert np.array_equal(I.collect(), np.rint(g).astype(np.int64))
+    s = dja.dsum(I)
+    assert s == int(np.rint(g).astype(np.int64).sum())
+    for d in (D, F, B, I):
+        d.close()
+
+
+@pytest.mark.timeout(420)
+def test_world2_cast(tmp_path):
+    _spawn(tmp_path, 2, "cast")

@@ -661,3 +661,21 @@ def test_device_props(dja):
     assert hbm.value > 200 * 1024 ** 3   # 288 GB HBM3E
     # note: device NAME can be empty on pool boxes (libdrm name quirk —
     # rocm-smi shows the same); only the memory size is asserted
+
+
+def test_dcast(dja):
+    """DArray{T2}(D) conversions: f64<->f32<->i64, half-even float->int."""
+    x = philox.fill_uniform_f64(30011, seed=40) * 100.0
+    d = dja.distribute(x)
+    f = dja.dcast(d, "f32")
+    assert np.array_equal(f.localpart(), x.astype(np.float32))
+    b = dja.dcast(f, "f64")
+    assert np.array_equal(b.localpart(),
+                          x.astype(np.float32).astype(np.float64))
+    i = dja.dcast(d, "i64")
+    assert np.array_equal(i.localpart(), np.rint(x).astype(np.int64))
+    g = dja.dcast(i, "f64")
+    assert np.array_equal(g.localpart(),
+                          np.rint(x).astype(np.int64).astype(np.float64))
+    for t in (d, f, b, i, g):
+        t.close()
