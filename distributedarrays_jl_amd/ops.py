@@ -83,7 +83,25 @@ def map2_(op, dest, a, b):
     return dest
 
 
+_PROMOTE_RANK = {"i64": 0, "f32": 1, "f64": 2}
+
+
 def elementwise(op, a, b):
+    """C = f.(a, b) out of place, with Julia's type promotion for mixed
+    eltypes (promote_type: Int64 < Float32 < Float64).  In-place map2_
+    keeps strict same-dtype semantics (the destination fixes T)."""
+    if a.dtype != b.dtype:
+        target = (a.dtype if _PROMOTE_RANK[a.dtype] >=
+                  _PROMOTE_RANK[b.dtype] else b.dtype)
+        ca = a if a.dtype == target else dcast(a, target)
+        cb = b if b.dtype == target else dcast(b, target)
+        try:
+            return map2_(op, ca.similar(), ca, cb)
+        finally:
+            if ca is not a:
+                ca.close()
+            if cb is not b:
+                cb.close()
     return map2_(op, a.similar(), a, b)
 
 

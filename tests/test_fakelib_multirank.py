@@ -612,3 +612,29 @@ def _scenario_cast(rank, world, dja):
 @pytest.mark.timeout(420)
 def test_world2_cast(tmp_path):
     _spawn(tmp_path, 2, "cast")
+
+
+def _scenario_promotion(rank, world, dja):
+    """Mixed-eltype elementwise promotes like Julia (i64<f32<f64)."""
+    from oracle import philox
+    n = 400
+    gf = _global_f64(n, 50)
+    gi = np.rint(_global_f64(n, 51) * 50).astype(np.int64)
+    D = _slice_set(dja.DArray((n,), "f64"), gf)
+    I = _slice_set(dja.DArray((n,), "i64"), gi)
+    S = dja.elementwise("add", D, I)
+    assert S.dtype == "f64"
+    assert np.array_equal(S.collect(), gf + gi.astype(np.float64))
+    g32 = philox.fill_uniform_f32(n, 52)
+    F = _slice_set(dja.DArray((n,), "f32"), g32)
+    M = dja.elementwise("mul", I, F)
+    assert M.dtype == "f32"
+    assert np.array_equal(M.collect(),
+                          gi.astype(np.float32) * g32)
+    for d in (D, I, S, F, M):
+        d.close()
+
+
+@pytest.mark.timeout(420)
+def test_world2_promotion(tmp_path):
+    _spawn(tmp_path, 2, "promotion")
