@@ -100,9 +100,10 @@ class Binary(Expr):
 
 
 def wrap(x):
+    import numbers
     if isinstance(x, Expr):
         return x
-    if isinstance(x, (int, float)):
+    if isinstance(x, numbers.Real):    # int/float incl. numpy scalars
         return Lit(x)
     # a bare DArray in an expression position
     if hasattr(x, "lidx") and hasattr(x, "dtype"):
@@ -160,9 +161,11 @@ def compile_expr(e):
                 args.append(node.d)
             prog.append((K_ARG << 8) | argids[key])
         elif isinstance(node, Lit):
-            if node.v in consts:
-                k = consts.index(node.v)
-            else:
+            import struct
+            key = struct.pack("<d", node.v)   # bit pattern: -0.0 != 0.0
+            k = next((i for i, c in enumerate(consts)
+                      if struct.pack("<d", c) == key), None)
+            if k is None:
                 if len(consts) >= MAXCONSTS:
                     raise DArrayError("expr: more than %d constants"
                                       % MAXCONSTS)
