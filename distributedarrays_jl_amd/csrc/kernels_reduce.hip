@@ -126,9 +126,13 @@ __device__ __forceinline__ T block_reduce(int redop, T v) {
     return v;
 }
 
-template <typename T>
+// MOP/ROP >= 0 fold the op switches at compile time for the hot
+// combos (same trick as map_fixed/dims-reduce); -1 = runtime args.
+template <typename T, int MOP = -1, int ROP = -1>
 __global__ void reduce_stage1(int mapop, int redop, const T* __restrict__ src,
                               uint64_t n, T* __restrict__ partials) {
+    if (MOP >= 0) mapop = MOP;
+    if (ROP >= 0) redop = ROP;
     uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
     uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
     T acc = RedIdent<T>::get(redop);
@@ -148,10 +152,12 @@ __global__ void reduce_stage1(int mapop, int redop, const T* __restrict__ src,
 
 // 32 B/thread-iteration variant (two dwordx4 per lane): more memory-
 // level parallelism per wave — A/B-gated via DA_RV4.
-template <typename T>
+template <typename T, int MOP = -1, int ROP = -1>
 __global__ void reduce_stage1_v4(int mapop, int redop,
                                  const T* __restrict__ src, uint64_t n,
                                  T* __restrict__ partials) {
+    if (MOP >= 0) mapop = MOP;
+    if (ROP >= 0) redop = ROP;
     uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
     uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
     T acc = RedIdent<T>::get(redop);
@@ -264,11 +270,31 @@ static int do_reduce(int mapop, int redop, const T* src, uint64_t n,
         DA_CHECK_HIP(hipGetLastError());
     } else {
         if (v4)
-            hipLaunchKernelGGL(reduce_stage1_v4<T>, dim3(g), dim3(RTPB), 0,
-                               s, mapop, redop, src, n, parts);
-        else
-            hipLaunchKernelGGL(reduce_stage1<T>, dim3(g), dim3(RTPB), 0, s,
+            hipLaunchKernelGGL((reduce_stage1_v4<T>), dim3(g), dim3(RTPB),
+                               0, s, mapop, redop, src, n, parts);
+        else if (mapop == DA_REDF_IDENTITY && redop == DA_RED_ADD)
+            hipLaunchKernelGGL((reduce_stage1<T, DA_REDF_IDENTITY,
+                                DA_RED_ADD>), dim3(g), dim3(RTPB), 0, s,
                                mapop, redop, src, n, parts);
+        else if (mapop == DA_REDF_ABS2 && redop == DA_RED_ADD)
+            hipLaunchKernelGGL((reduce_stage1<T, DA_REDF_ABS2,
+                                DA_RED_ADD>), dim3(g), dim3(RTPB), 0, s,
+                               mapop, redop, src, n, parts);
+        else if (mapop == DA_REDF_ABS && redop == DA_RED_ADD)
+            hipLaunchKernelGGL((reduce_stage1<T, DA_REDF_ABS,
+                                DA_RED_ADD>), dim3(g), dim3(RTPB), 0, s,
+                               mapop, redop, src, n, parts);
+        else if (mapop == DA_REDF_IDENTITY && redop == DA_RED_MAX)
+            hipLaunchKernelGGL((reduce_stage1<T, DA_REDF_IDENTITY,
+                                DA_RED_MAX>), dim3(g), dim3(RTPB), 0, s,
+                               mapop, redop, src, n, parts);
+        else if (mapop == DA_REDF_IDENTITY && redop == DA_RED_MIN)
+            hipLaunchKernelGGL((reduce_stage1<T, DA_REDF_IDENTITY,
+                                DA_RED_MIN>), dim3(g), dim3(RTPB), 0, s,
+                               mapop, redop, src, n, parts);
+        else
+            hipLaunchKernelGGL((reduce_stage1<T>), dim3(g), dim3(RTPB),
+                               0, s, mapop, redop, src, n, parts);
         DA_CHECK_HIP(hipGetLastError());
         hipLaunchKernelGGL(reduce_stage2<T>, dim3(1), dim3(RTPB), 0, s,
                            redop, parts, g, dout);
