@@ -151,3 +151,29 @@ def test_fullsize_expr_property(dja):
     assert abs(s - ((1 - np.cos(1.0)) + 0.25)) < 1e-3
     for d in (A, B, D):
         d.close()
+
+
+def test_chunk_beyond_4gib(dja):
+    """A single chunk larger than 2^32 BYTES (4.5 GiB): u64 indexing
+    end-to-end through rand/map/expr/reduce (would catch any i32
+    offset truncation)."""
+    from distributedarrays_jl_amd import expr as E
+    n = (1 << 29) + 12345          # 4.295e9 B of f64
+    D = dja.drand((n,), "f64")
+    rng = np.random.default_rng(5)
+    # spot-check philox elements across the whole range incl. the tail
+    idxs = sorted(set(int(i) for i in rng.integers(0, n, 64))
+                  | {0, n - 1, (1 << 29) - 1, 1 << 29})
+    got = _read_elems(D, idxs)
+    assert np.array_equal(got, _philox_at(1234, np.array(idxs)))
+    s = dja.dsum(D) / n
+    assert abs(s - 0.5) < 1e-4
+    dja.map_("sin", D, D)
+    got2 = _read_elems(D, idxs)
+    assert np.allclose(got2, np.sin(_philox_at(1234, np.array(idxs))),
+                       rtol=1e-14, atol=1e-15)
+    O = E.materialize(E.ref(D) * 2.0 - 0.25)
+    got3 = _read_elems(O, idxs)
+    assert np.array_equal(
+        got3, np.sin(_philox_at(1234, np.array(idxs))) * 2.0 - 0.25)
+    D.close(); O.close()
