@@ -20,7 +20,7 @@ sequence of single-op kernels."""
 import ctypes
 
 from ._ffi import check, lib, DArrayError
-from ._opcodes import (DTYPES, DTYPE_SIZE, MAP_OP, MAP2_OP,
+from ._opcodes import (DTYPES, MAP_OP, MAP2_OP,
                        I64_MAP_OPS, I64_MAP2_OPS)
 
 K_UNARY, K_ARG, K_CONST, K_BINARY = 0, 1, 2, 3

@@ -9,7 +9,7 @@ becomes an RCCL allreduce (SURVEY.md §3.1 build mapping)."""
 import ctypes
 import numpy as np
 
-from . import comm, geometry, plan
+from . import geometry, plan
 from ._ffi import check, lib, DArrayError
 from ._opcodes import (DTYPES, DTYPE_SIZE, NUMPY_DTYPES, MAP_OP, MAP2_OP,
                        RED_OPS, RED_FS, I64_MAP_OPS, I64_MAP2_OPS)

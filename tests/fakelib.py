@@ -27,8 +27,7 @@ import numpy as np
 
 import oracle.ops as oops
 import oracle.philox as ophilox
-from distributedarrays_jl_amd._opcodes import (
-    MAP_OPS, MAP2_OPS, DTYPE_SIZE)
+from distributedarrays_jl_amd._opcodes import MAP_OPS, MAP2_OPS
 
 _NPDT = {0: np.dtype("float64"), 1: np.dtype("float32"),
          2: np.dtype("int64")}
