@@ -174,6 +174,8 @@ def test_chunk_beyond_4gib(dja):
                        rtol=1e-14, atol=1e-15)
     O = E.materialize(E.ref(D) * 2.0 - 0.25)
     got3 = _read_elems(O, idxs)
-    assert np.array_equal(
-        got3, np.sin(_philox_at(1234, np.array(idxs))) * 2.0 - 0.25)
+    # compare against D's ACTUAL values (got2): the chain *2 - 0.25 is
+    # exact arithmetic, but D's sin differs from libm's by <=1 ulp, so
+    # anchoring on np.sin would demand cross-library bit-equality
+    assert np.array_equal(got3, got2 * 2.0 - 0.25)
     D.close(); O.close()
