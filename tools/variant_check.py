@@ -24,6 +24,9 @@ assert np.abs(C.localpart() - ref).max() / np.abs(ref).max() < 1e-12
 x = philox.fill_uniform_f64(100001, 3)
 d = dja.distribute(x)
 assert abs(dja.dsum(d) - x.sum()) < 1e-6
+from distributedarrays_jl_amd import expr as E
+o = E.materialize(E.ref(d) * 2.0 + 1.0)
+assert np.array_equal(o.localpart(), x * 2.0 + 1.0)
 print("variant ok:", {k: v for k, v in os.environ.items() if k.startswith("DA_")})
 '''
 
@@ -32,6 +35,7 @@ VARIANTS = [
     {"DA_GEMM_V": "5"}, {"DA_GEMM_V": "2", "DA_GEMM_BK": "32"},
     {"DA_RED_FUSED": "1"}, {"DA_NT": "1"}, {"DA_RV4": "1"},
     {"DA_RBLOCKS": "4096"}, {"DA_MM_OVERLAP": "0"},
+    {"DA_EXPR_JIT": "0"},
 ]
 
 
