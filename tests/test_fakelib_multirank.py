@@ -576,6 +576,36 @@ def test_world4(tmp_path, name):
     _spawn(tmp_path, 4, name)
 
 
+def _scenario_random_ragged(rank, world, dja):
+    """Random ragged layouts (from_chunk_sizes) through the transparent
+    routing paths: every op pair below crosses mismatched cut vectors,
+    including empty chunks."""
+    rng = np.random.default_rng(424242)
+    for it in range(8):
+        n = int(rng.integers(1, 300))
+        # random composition of n into `world` parts (empties allowed)
+        cuts = np.sort(rng.integers(0, n + 1, world - 1))
+        sizes = np.diff(np.concatenate([[0], cuts, [n]])).tolist()
+        assert sum(sizes) == n and len(sizes) == world
+        ga = _global_f64(n, 5000 + it)
+        gb = _global_f64(n, 6000 + it)
+        A = _slice_set(dja.DArray((n,), "f64"), ga)
+        R = dja.DArray.from_chunk_sizes(sizes, "f64")
+        _slice_set(R, gb)
+        out = dja.elementwise("mul", A, R)
+        assert np.allclose(out.collect(), ga * gb, rtol=0), (it, sizes)
+        out2 = dja.DArray.from_chunk_sizes(sizes, "f64")
+        dja.map2_("sub", out2, R, A)
+        assert np.allclose(out2.collect(), gb - ga, rtol=0), (it, sizes)
+        got = dja.ddot(A, R)
+        assert abs(got - float(ga @ gb)) <= 1e-12 * max(
+            abs(float(ga @ gb)), 1e-30), (it, sizes)
+        s = dja.dsum(R)
+        assert abs(s - gb.sum()) <= 1e-12 * max(abs(gb.sum()), 1e-30)
+        for d in (A, R, out, out2):
+            d.close()
+
+
 def _scenario_random_sweep_b(rank, world, dja):
     _scenario_random_sweep(rank, world, dja, seed=31337)
 
@@ -585,7 +615,8 @@ def _scenario_random_sweep_c(rank, world, dja):
 
 
 @pytest.mark.timeout(420)
-@pytest.mark.parametrize("name", ["random_sweep_b", "random_sweep_c"])
+@pytest.mark.parametrize("name", ["random_sweep_b", "random_sweep_c",
+                                  "random_ragged"])
 def test_world4_sweep_seeds(tmp_path, name):
     _spawn(tmp_path, 4, name)
 
