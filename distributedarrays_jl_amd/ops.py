@@ -1078,24 +1078,31 @@ def dppeval(f, *Ds, dim=None):
     if not Ds or not isinstance(Ds[0], DArray):
         raise DArrayError("dppeval: first argument must be a DArray "
                           "(procs(D[1]) in the reference)")
-    dax = []
-    for a in Ds:
-        dax.append(a.ndims - 1 if isinstance(a, DArray) else 0)
-    if dim is not None:
-        dax = [d - 1 if d > 0 else 0 for d in dim]   # 1-based like ref
+    # dim entries are 1-based like the reference; <= 0 means the
+    # argument is broadcast whole to every evaluation (mapreduce.jl
+    # ppeval docstring)
+    if dim is None:
+        dax = [a.ndims - 1 if isinstance(a, DArray) else -1 for a in Ds]
+        sliced = [isinstance(a, DArray) for a in Ds]
+    else:
+        dax = [d - 1 for d in dim]
+        sliced = [isinstance(a, DArray) and d > 0
+                  for a, d in zip(Ds, dim)]
+    if not sliced[0]:
+        raise DArrayError("dppeval: the first DArray must be sliced")
     first = Ds[0]
-    for a, d in zip(Ds, dax):
+    for a, d, sl in zip(Ds, dax, sliced):
         if isinstance(a, DArray):
             for ax in range(a.ndims):
-                if ax != d and a.dist[ax] != 1:
+                if (ax != d or not sl) and a.dist[ax] != 1:
                     raise DArrayError(
                         "dppeval: dimension %d is distributed; must be "
                         "whole per chunk" % ax)
     locals_ = [a.localpart() if isinstance(a, DArray) else np.asarray(a)
                for a in Ds]
 
-    def slice_at(arr, d, i, is_d):
-        if not is_d:
+    def slice_at(arr, d, i, is_sliced):
+        if not is_sliced:
             return arr
         idx = [slice(None)] * arr.ndim
         idx[d] = i
@@ -1104,8 +1111,8 @@ def dppeval(f, *Ds, dim=None):
     n_loc = locals_[0].shape[dax[0]]
     outs = []
     for i in range(n_loc):
-        args = [slice_at(lp, d, i, isinstance(a, DArray))
-                for lp, d, a in zip(locals_, dax, Ds)]
+        args = [slice_at(lp, d, i, sl)
+                for lp, d, sl in zip(locals_, dax, sliced)]
         outs.append(np.asarray(f(*args)))
     if outs:
         local = np.stack(outs, axis=-1)
