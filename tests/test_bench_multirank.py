@@ -72,6 +72,11 @@ def _spawn(tmp_path, world):
 
 
 @pytest.mark.timeout(600)
+def test_bench_world2(tmp_path):
+    _spawn(tmp_path, 2)
+
+
+@pytest.mark.timeout(600)
 def test_bench_world4_cfg3_branch(tmp_path):
     _spawn(tmp_path, 4)
 
