@@ -34,6 +34,12 @@ Parity pinning status: PINNED.
   inputs) with the exactness split the reference's tests use
   (/root/reference/test/darray.jl:286-294,439-452: exact for integers,
   `sqrt(eps())`-style tolerance for float linalg).
+- Broadcast-composition semantics (oracle/expr.py, round 2) restate the
+  Broadcasted-tree materialization of /root/reference/src/broadcast.jl:65-98
+  over the same postfix encoding the product ships to da_expr; nested
+  broadcast behaviour is pinned by /root/reference/test/darray.jl:880-912
+  (mirrored in tests/test_gpu_expr.py) and the encoding tables are pinned
+  to the C header via tests/test_expr.py + tests/test_abi.py.
 """
 
 from .philox import (philox4x32, fill_uniform_f64, fill_uniform_f32,
