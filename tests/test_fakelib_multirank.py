@@ -678,19 +678,13 @@ def _scenario_ppeval_broadcast_darray(rank, world, dja):
     A3 = dja.DArray((3, 3, nslices), "f64", (1, 1, world))
     g3 = _global_f64(9 * nslices, 60).reshape((3, 3, nslices), order="F")
     _slice_set(A3, g3)
-    # W replicated per rank as a fully-undistributed DArray
+    # the broadcast-whole argument: a non-DArray (host) array with
+    # dim=0, the reference's "not distributed -> broadcast" clause
     gw = _global_f64(9, 61).reshape((3, 3), order="F")
-    W = dja.DArray((3, 3), "f64", (1, 1))
-    if W.lnumel:
-        W.set_localpart(np.asfortranarray(gw))
-    # note: W's chunk lives on rank 0 only; broadcast-whole semantics
-    # need the local value, so pass the host array instead on other
-    # ranks — the portable broadcast form is a non-DArray argument,
-    # matching the reference's "not distributed -> broadcast" clause
     P = dja.dppeval(lambda s, b: s @ b, A3, gw, dim=(3, 0))
     ref = np.stack([g3[:, :, i] @ gw for i in range(nslices)], axis=-1)
     assert np.allclose(P.collect(), ref, rtol=1e-12)
-    P.close(); A3.close(); W.close()
+    P.close(); A3.close()
 
 
 @pytest.mark.timeout(420)
