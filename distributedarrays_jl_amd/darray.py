@@ -333,6 +333,49 @@ class DArray:
                              DTYPE_SIZE[self.dtype]))
         return value
 
+    def __getitem__(self, key):
+        """D[i, j] / D[a:b, c:d] — the reference's scalar and
+        contiguous-range getindex (darray.jl:637-820, within this
+        build's range scope; strided/fancy indexing is out of scope).
+        Collective at nranks>1: every rank must index identically.
+        Scalars gate on allowscalar; ranges gather via the makelocal
+        box path and return a numpy array on every rank."""
+        if not isinstance(key, tuple):
+            key = (key,)
+        if len(key) != self.ndims:
+            raise _ffi.DArrayError(
+                "indexing needs %d subscripts" % self.ndims)
+        if all(isinstance(k, (int, np.integer)) for k in key):
+            return self.getindex(*(int(k) for k in key))
+        from . import ops
+        ranges = []
+        for d, k in enumerate(key):
+            if isinstance(k, (int, np.integer)):
+                ranges.append((int(k), int(k) + 1))
+            elif isinstance(k, slice):
+                if k.step not in (None, 1):
+                    raise _ffi.DArrayError(
+                        "strided ranges are out of scope")
+                lo = 0 if k.start is None else int(k.start)
+                hi = self.dims[d] if k.stop is None else int(k.stop)
+                ranges.append((lo, hi))
+            else:
+                raise _ffi.DArrayError("bad subscript %r" % (k,))
+        out = ops.dgetindex(self, *ranges)
+        squeeze = tuple(d for d, k in enumerate(key)
+                        if isinstance(k, (int, np.integer)))
+        return out.squeeze(axis=squeeze) if squeeze else out
+
+    def __setitem__(self, key, value):
+        """Scalar D[i, j] = v (darray.jl:700ff); collective."""
+        if not isinstance(key, tuple):
+            key = (key,)
+        if not all(isinstance(k, (int, np.integer)) for k in key):
+            raise _ffi.DArrayError(
+                "only scalar setindex is supported (range assignment "
+                "is the out-of-scope SubDArray machinery)")
+        self.setindex(value, *(int(k) for k in key))
+
     def __eq__(self, other):
         """== is elementwise-all equality (test/darray.jl:84-129);
         mismatched cuts localize `other` onto self's boxes via the

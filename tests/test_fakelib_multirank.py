@@ -551,7 +551,18 @@ def _scenario_scalar_index(rank, world, dja):
     assert D.getindex(n - 1) == g[n - 1]
     D.setindex(3.25, 2)
     assert D.getindex(2) == 3.25
-    D.close()
+    # __getitem__/__setitem__ sugar (scalar + contiguous ranges)
+    assert D[5] == g[5]
+    assert np.array_equal(D[10:20], g[10:20])
+    D[7] = -1.5
+    assert D[7] == -1.5
+    g2 = _global_f64(48, 18).reshape((8, 6), order="F")
+    M = _slice_set(dja.DArray((8, 6), "f64"), g2)
+    assert M[3, 4] == g2[3, 4]
+    assert np.array_equal(M[2:6, 1:5], g2[2:6, 1:5])
+    assert np.array_equal(M[3, 1:5], g2[3, 1:5])
+    assert np.array_equal(M[:, 2], g2[:, 2])
+    D.close(); M.close()
 
 
 # ------------------------------------------------------------- test entry
