@@ -679,3 +679,23 @@ def test_dcast(dja):
                           np.rint(x).astype(np.int64).astype(np.float64))
     for t in (d, f, b, i, g):
         t.close()
+
+
+def test_index_sugar(dja):
+    """__getitem__/__setitem__: scalar + contiguous ranges
+    (darray.jl:637-820 in-scope subset)."""
+    x = philox.fill_uniform_f64(500, seed=50)
+    d = dja.distribute(x)
+    assert d[7] == x[7]
+    assert np.array_equal(d[100:200], x[100:200])
+    d[3] = 9.75
+    assert d[3] == 9.75
+    d.close()
+    m = np.asfortranarray(philox.fill_uniform_f64(12 * 8, 51)
+                          .reshape(12, 8, order="F"))
+    M = dja.distribute(m)
+    assert M[5, 6] == m[5, 6]
+    assert np.array_equal(M[2:9, 3:7], m[2:9, 3:7])
+    assert np.array_equal(M[4, 0:8], m[4, 0:8])
+    assert np.array_equal(M[:, 5], m[:, 5])
+    M.close()
