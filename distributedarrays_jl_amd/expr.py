@@ -321,8 +321,12 @@ def materialize(e):
         if a.dims == dims:
             proto = a
             break
-    if proto is None:
-        raise DArrayError("expr: no operand has the full broadcast "
-                          "shape %r" % (dims,))
-    dest = proto.similar()
+    if proto is not None:
+        dest = proto.similar()
+    else:
+        # outer-product broadcasting (e.g. row .+ col -> (m, n)): no
+        # operand carries the result shape, so allocate it on the
+        # default grid; every operand localizes via projected boxes
+        from .darray import DArray
+        dest = DArray(dims, args[0].dtype)
     return materialize_(dest, e)

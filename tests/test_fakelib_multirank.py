@@ -701,3 +701,23 @@ def _scenario_ppeval_broadcast_darray(rank, world, dja):
 @pytest.mark.timeout(420)
 def test_world2_ppeval_broadcast(tmp_path):
     _spawn(tmp_path, 2, "ppeval_broadcast_darray")
+
+
+def _scenario_outer_broadcast(rank, world, dja):
+    """Outer-product broadcasting: row .* col -> (m, n) with NO
+    full-shape operand (both localize via projected boxes)."""
+    from distributedarrays_jl_amd import expr as E
+    m, n = 10, 4 * world
+    grow = _global_f64(n, 70).reshape((1, n), order="F")
+    gcol = _global_f64(m, 71).reshape((m, 1), order="F")
+    R = _slice_set(dja.DArray((1, n), "f64", (1, world)), grow)
+    C = _slice_set(dja.DArray((m, 1), "f64", (1, 1)), gcol)
+    O = E.materialize(E.ref(C) * E.ref(R) + 1.0)
+    assert O.dims == (m, n)
+    assert np.allclose(O.collect(), gcol * grow + 1.0, rtol=0)
+    O.close(); R.close(); C.close()
+
+
+@pytest.mark.timeout(420)
+def test_world2_outer_broadcast(tmp_path):
+    _spawn(tmp_path, 2, "outer_broadcast")

@@ -210,3 +210,17 @@ def test_validation_errors(dja):
     with pytest.raises(DArrayError):
         E.materialize_(d, E.ref(bad) + 1.0)
     d.close(); bad.close()
+
+
+def test_outer_product_broadcast(dja):
+    """row .* col -> (m, n): no full-shape operand (both stride-0
+    expand from projected boxes)."""
+    m, n = 48, 32
+    grow = philox.fill_uniform_f64(n, 30).reshape((1, n), order="F")
+    gcol = philox.fill_uniform_f64(m, 31).reshape((m, 1), order="F")
+    R = _mk(dja, grow)
+    C = _mk(dja, gcol)
+    O = E.materialize(E.ref(C) * E.ref(R) + 1.0)
+    assert O.dims == (m, n)
+    assert np.array_equal(O.localpart(), gcol * grow + 1.0)
+    O.close(); R.close(); C.close()
