@@ -154,3 +154,79 @@ def test_shape_validation():
     ok = FakeD((1, 4))
     prog, args, consts = E.compile_expr(E.ref(ok) + 1.0)
     E._validate(d, prog, args, consts)
+
+
+# ---------------------------------------------------------- properties
+from hypothesis import given, settings, strategies as st
+
+
+class _D(FakeD):
+    """Leaf stand-ins paired with host arrays for direct evaluation."""
+
+    def __init__(self, arr):
+        super().__init__(arr.shape)
+        self.arr = arr
+
+
+_PROP_UNARY = ["neg", "abs", "abs2", "sqrt", "inv", "floor", "sign"]
+_PROP_BINARY = ["add", "sub", "mul", "div", "min2", "max2"]
+
+
+def _tree_strategy(leaves):
+    leaf = st.one_of(
+        st.sampled_from([("ref", l) for l in leaves]),
+        st.floats(-4, 4, allow_nan=False).map(
+            lambda v: ("lit", round(v, 3))))
+    return st.recursive(
+        leaf,
+        lambda children: st.one_of(
+            st.tuples(st.sampled_from(_PROP_UNARY), children),
+            st.tuples(st.sampled_from(_PROP_BINARY), children, children)),
+        max_leaves=12)
+
+
+def _build(E, spec):
+    if spec[0] == "ref":
+        return E.ref(spec[1])
+    if spec[0] == "lit":
+        return E.lit(spec[1])
+    if len(spec) == 2:
+        return getattr(E, spec[0])(_build(E, spec[1]))
+    from distributedarrays_jl_amd.expr import Binary
+    return Binary(spec[0], _build(E, spec[1]), _build(E, spec[2]))
+
+
+def _direct(spec):
+    """Independent tree-walking numpy evaluation (no postfix)."""
+    import oracle.ops as oops
+    if spec[0] == "ref":
+        return spec[1].arr
+    if spec[0] == "lit":
+        return np.float64(spec[1])
+    if len(spec) == 2:
+        return oops.MAP_OPS[spec[0]](np.asarray(_direct(spec[1]),
+                                                dtype=np.float64))
+    return oops.MAP2_OPS[spec[0]](_direct(spec[1]), _direct(spec[2]))
+
+
+@given(st.data())
+@settings(max_examples=150, deadline=None)
+def test_compile_eval_equals_direct_tree_eval(data):
+    """compile_expr -> postfix evaluate must equal an independent
+    direct walk of the same tree (pins the encoding end-to-end)."""
+    from distributedarrays_jl_amd import expr as E
+    leaves = [_D(philox.fill_uniform_f64(37, 800 + k) + 0.25)
+              for k in range(3)]
+    spec = data.draw(_tree_strategy(leaves))
+    try:
+        prog, args, consts = E.compile_expr(_build(E, spec))
+    except DArrayError:
+        return          # over the documented limits — fine
+    with np.errstate(all="ignore"):
+        got = oexpr.evaluate(prog, [a.arr for a in args], consts)
+        want = _direct(spec)
+    got = np.broadcast_to(np.asarray(got, dtype=np.float64), (37,)) \
+        if np.ndim(got) else np.full(37, got)
+    want = np.broadcast_to(np.asarray(want, dtype=np.float64), (37,)) \
+        if np.ndim(want) else np.full(37, want)
+    assert np.array_equal(got, want, equal_nan=True), spec
