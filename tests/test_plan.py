@@ -117,3 +117,34 @@ def test_partial_plan_per_k_partition():
         assert sorted(parts) == sorted(moves)
         # ascending-k issue order is identical on every rank by
         # construction (k loop), so pairing matches the single group
+
+
+def test_partial_exchange_skip_pairing():
+    """The zero-size skip conditions in ops.dmatmul's partial exchange
+    pair consistently: the sender's byte count (A row-block height x
+    C column-block width) IS the receiver's C.lnumel, so both sides
+    drop exactly the same moves — over random geometries including
+    degenerate cuts (sz < chunks)."""
+    import numpy as np
+    from distributedarrays_jl_amd import geometry as pg, plan
+    rng = np.random.default_rng(7)
+    for _ in range(200):
+        m = int(rng.integers(1, 20))
+        n = int(rng.integers(1, 20))
+        I = int(rng.integers(1, 4))
+        J = int(rng.integers(1, 4))
+        K = int(rng.integers(1, J + 1))
+        rrows = pg.ranges1d(pg.cuts1d(m, I))
+        ccols = pg.ranges1d(pg.cuts1d(n, K))
+        moves = plan.partial_plan((I, J), K)
+        for (src, dst, k) in moves:
+            i_src = src % I
+            i_dst = dst % I
+            assert i_src == i_dst          # partials move within a row
+            send_nb = (rrows[i_src][1] - rrows[i_src][0]) * \
+                (ccols[k][1] - ccols[k][0])
+            kk_dst = dst // I
+            assert kk_dst == k             # to the k-th column owner
+            recv_nb = (rrows[i_dst][1] - rrows[i_dst][0]) * \
+                (ccols[kk_dst][1] - ccols[kk_dst][0])
+            assert send_nb == recv_nb      # skip iff both skip
