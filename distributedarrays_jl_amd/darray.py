@@ -227,10 +227,25 @@ class DArray:
         return self
 
     def similar(self, dtype=None):
-        # propagate the chunk-owner mapping (non-identity for e.g.
-        # dims-reduction results) so elementwise ops stay aligned
-        return DArray(self.dims, dtype or self.dtype, self.dist,
-                      ranks=list(self.ranks))
+        """A new uninitialized DArray with THIS array's exact layout:
+        same cut boxes (incl. ragged from_chunk_sizes cuts) and the
+        same chunk->rank owners — so elementwise ops on the result stay
+        aligned and local (Julia's similar preserves the
+        distribution)."""
+        d = DArray(self.dims, dtype or self.dtype, self.dist,
+                   ranks=list(self.ranks), _alloc=False)
+        d.cuts = [list(c) for c in self.cuts]
+        d.idxs = list(self.idxs)
+        d.lchunk = self.lchunk
+        d.lidx = self.lidx
+        d.lshape = self.lshape
+        d.lnumel = self.lnumel
+        p = ctypes.c_void_p()
+        check(lib.da_alloc(max(d.lnumel, 1) * DTYPE_SIZE[d.dtype],
+                           DTYPES[d.dtype], ctypes.byref(p)))
+        d._chunk = p
+        _registry[d.id] = d
+        return d
 
     def copy(self):
         out = self.similar()

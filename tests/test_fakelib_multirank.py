@@ -141,7 +141,15 @@ def _scenario_routing(rank, world, dja):
     M = dja.DArray.from_chunk_sizes(sizes, "f64")
     dja.map_("abs2", M, A)
     assert np.allclose(M.collect(), ga * ga, rtol=0)
-    for d in (A, R, Dst, Dst2, A2, F, S, Rcopy, M):
+    # similar()/dmap on a ragged array PRESERVES the ragged layout
+    # (stays aligned and local; Julia's similar keeps the distribution)
+    Rm = dja.dmap("neg", R)
+    assert Rm.samedist(R) and R.samedist(Rm)
+    assert np.allclose(Rm.collect(), -gr, rtol=0)
+    Rc = R.copy()
+    assert Rc.samedist(R)
+    assert np.array_equal(Rc.collect(), gr)
+    for d in (A, R, Dst, Dst2, A2, F, S, Rcopy, M, Rm, Rc):
         d.close()
 
 
