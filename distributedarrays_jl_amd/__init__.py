@@ -19,7 +19,8 @@ from ._ffi import DArrayError
 from . import comm, geometry, plan, spmd, expr
 from .darray import (DArray, dzeros, dones, dfill, drand, drandn,
                      distribute, localpart, localindices, d_closeall,
-                     bytes_in_use, ddata, dgather, locate, allowscalar)
+                     bytes_in_use, ddata, dgather, locate, allowscalar,
+                     dfromfunction)
 from .ops import (map_, dmap, map2_, elementwise, map2_scalar_, elementwise_scalar, broadcast_fma, axpy_,
                   add_, scale_, mapreduce, dsum, dprod, dmaximum, dminimum,
                   dextrema, dmean, dcount, dall, dany, ddot, dnorm, dmatmul, dreduce_dims,
@@ -35,7 +36,7 @@ __all__ = [
     "DArray", "DArrayError", "comm", "geometry", "plan", "spmd", "expr",
     "dzeros", "dones", "dfill", "drand", "drandn", "distribute",
     "localpart", "localindices", "d_closeall", "bytes_in_use",
-    "ddata", "dgather", "locate", "allowscalar",
+    "ddata", "dgather", "locate", "allowscalar", "dfromfunction",
     "map_", "dmap", "map2_", "elementwise", "map2_scalar_", "elementwise_scalar", "broadcast_fma", "axpy_",
     "add_", "scale_", "mapreduce", "dsum", "dprod", "dmaximum",
     "dminimum", "dextrema", "dmean", "dcount", "dall", "dany", "ddot", "dnorm", "dmatmul",

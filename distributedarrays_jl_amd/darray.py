@@ -469,6 +469,18 @@ def distribute(a, dist=None):
                   init=lambda idx: a[tuple(slice(lo, hi) for lo, hi in idx)])
 
 
+def dfromfunction(f, dims, dtype="f64", dist=None):
+    """@DArray [f(i, j, ...) for i=..., j=...] — the comprehension
+    sugar of the reference's docs (docs/src/index.md; `@DArray` macro,
+    darray.jl:214-231): each rank builds its chunk by evaluating f over
+    its GLOBAL index ranges (0-based here, numpy fromfunction style)."""
+    def init(idx):
+        return np.asfortranarray(np.fromfunction(
+            lambda *loc: f(*[l + lo for l, (lo, _) in zip(loc, idx)]),
+            geometry.shape_of(idx)))
+    return DArray(dims, dtype, dist, init=init)
+
+
 def localpart(d):
     return d.localpart()
 

@@ -93,7 +93,11 @@ def _scenario_basic(rank, world, dja):
     g = _global_f64(n, seed=42)
     E = dja.distribute(g)
     assert np.array_equal(E.collect(), g)
-    for d in (D, M, P, E):
+    # @DArray comprehension analog: f over global indices
+    F = dja.dfromfunction(lambda i, j: i + j, (5, 8))
+    ref = np.add.outer(np.arange(5), np.arange(8)).astype(np.float64)
+    assert np.array_equal(F.collect(), ref)
+    for d in (D, M, P, E, F):
         d.close()
 
 
