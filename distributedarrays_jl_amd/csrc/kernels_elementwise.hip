@@ -258,6 +258,39 @@ __global__ void map_fixed_kernel(T* __restrict__ dst,
         dst[j] = apply_map<T>(OP, src[j]);
 }
 
+// 32 B/lane variant (double4): more memory-level parallelism per
+// wave iteration — A/B-gated via DA_MAP_W=4 against the 16 B default.
+template <typename T, int OP>
+__global__ void map_fixed_kernel_w4(T* __restrict__ dst,
+                                    const T* __restrict__ src, uint64_t n) {
+    uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    using V = T __attribute__((ext_vector_type(4)));
+    uint64_t nv = n / 4;
+    const V* sv = reinterpret_cast<const V*>(src);
+    V* dv = reinterpret_cast<V*>(dst);
+    for (uint64_t j = i; j < nv; j += stride) {
+        V v = sv[j];
+        V r;
+        r.x = apply_map<T>(OP, v.x);
+        r.y = apply_map<T>(OP, v.y);
+        r.z = apply_map<T>(OP, v.z);
+        r.w = apply_map<T>(OP, v.w);
+        dv[j] = r;
+    }
+    for (uint64_t j = 4 * nv + i; j < n; j += stride)
+        dst[j] = apply_map<T>(OP, src[j]);
+}
+
+static inline int map_width() {
+    static int w = -1;
+    if (w < 0) {
+        const char* e = getenv("DA_MAP_W");
+        w = e ? atoi(e) : 2;
+    }
+    return w;
+}
+
 #define DA_HOT_OPS(X)                                                   \
     X(DA_OP_IDENTITY) X(DA_OP_NEG) X(DA_OP_ABS) X(DA_OP_ABS2)           \
     X(DA_OP_INV) X(DA_OP_SQRT) X(DA_OP_EXP) X(DA_OP_LOG)                \
@@ -270,8 +303,12 @@ static bool launch_map_hot(int opcode, T* dst, const T* src, uint64_t n,
     switch (opcode) {
 #define DA_CASE(OPC)                                                    \
     case OPC:                                                           \
-        hipLaunchKernelGGL((map_fixed_kernel<T, OPC>), dim3(g),         \
-                           dim3(TPB), 0, s, dst, src, n);               \
+        if (map_width() == 4)                                           \
+            hipLaunchKernelGGL((map_fixed_kernel_w4<T, OPC>), dim3(g),  \
+                               dim3(TPB), 0, s, dst, src, n);           \
+        else                                                            \
+            hipLaunchKernelGGL((map_fixed_kernel<T, OPC>), dim3(g),     \
+                               dim3(TPB), 0, s, dst, src, n);           \
         return true;
     DA_HOT_OPS(DA_CASE)
 #undef DA_CASE

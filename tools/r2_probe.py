@@ -63,6 +63,24 @@ def main():
         out("map_sin_nt1", t * 1e3, n * 16 / t / 1e9)
         return
 
+    if "width" in legs:
+        # structural gap probe: map identity vs raw D2D copy, W=2 vs 4
+        D = dja.drand((n,), "f64")
+        S = dja.DArray((n,), "f64")
+        t = bench(lambda: dja.map_("identity", S, D))
+        out("map_identity", t * 1e3, n * 16 / t / 1e9)
+        t = bench(lambda: dja.map_("sin", S, D))
+        out("map_sin_w", t * 1e3, n * 16 / t / 1e9)
+        t = bench(lambda: check(lib.da_d2d(S._ptr(), D._ptr(), n * 8)))
+        out("d2d_copy", t * 1e3, n * 16 / t / 1e9,
+            {"note": "hipMemcpyAsync DtoD; 16 B/elem bus (R+W)"})
+        A = dja.drand((n,), "f64")
+        T2 = dja.DArray((n,), "f64")
+        t = bench(lambda: dja.broadcast_fma(T2, D, A, 0.5))
+        out("bcast_fma_w", t * 1e3, n * 24 / t / 1e9)
+        for d in (D, S, A, T2):
+            d.close()
+
     if "sin" in legs:
         D = dja.drand((n,), "f64")
         S = dja.DArray((n,), "f64")
