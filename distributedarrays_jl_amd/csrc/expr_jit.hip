@@ -132,16 +132,19 @@ std::string gen_source(const int32_t* prog, int plen, int dtype, int nd,
             "  unsigned long long st = (unsigned long long)gridDim.x * "
             "blockDim.x;\n"
             "  %s* dst = (%s*)a.dst;\n"
-            "  unsigned long long nv = a.n / 2;\n"
+            "  unsigned long long nv = a.n / 4;\n"
             "  for (unsigned long long jp = i; jp < nv; jp += st) {\n"
-            "    unsigned long long j = 2 * jp;\n"
+            "    unsigned long long j = 4 * jp;\n"
             "    %s r0 = evalx(a, j);\n"
             "    %s r1 = evalx(a, j + 1);\n"
+            "    %s r2 = evalx(a, j + 2);\n"
+            "    %s r3 = evalx(a, j + 3);\n"
             "    dst[j] = r0; dst[j + 1] = r1;\n"
+            "    dst[j + 2] = r2; dst[j + 3] = r3;\n"
             "  }\n"
-            "  for (unsigned long long j = 2 * nv + i; j < a.n; j += st)\n"
+            "  for (unsigned long long j = 4 * nv + i; j < a.n; j += st)\n"
             "    dst[j] = evalx(a, j);\n"
-            "}\n", tname, tname, tname, tname);
+            "}\n", tname, tname, tname, tname, tname, tname);
         s += buf;
     } else {
         snprintf(buf, sizeof(buf),
@@ -275,7 +278,7 @@ int launch_expr_jit(const int32_t* prog, int plen, void* dst,
     void* cfg[] = {HIP_LAUNCH_PARAM_BUFFER_POINTER, &a,
                    HIP_LAUNCH_PARAM_BUFFER_SIZE, &asz,
                    HIP_LAUNCH_PARAM_END};
-    uint64_t work = strided ? n : n / 2 + 1;
+    uint64_t work = strided ? n : n / 4 + 1;
     uint64_t b = (work + 255) / 256;
     if (b > 8192) b = 8192;   // 1024 workgroups/XCD fills the chip
     if (b == 0) b = 1;
