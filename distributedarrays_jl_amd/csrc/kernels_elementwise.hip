@@ -551,7 +551,7 @@ static inline int bc_width() {
     static int w = -1;
     if (w < 0) {
         const char* e = getenv("DA_BC_W");
-        w = e ? atoi(e) : 2;
+        w = e ? atoi(e) : 4;   // measured 1.21 vs 1.37 ms on 2^28
     }
     return w;
 }
