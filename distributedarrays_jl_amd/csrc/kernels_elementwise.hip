@@ -626,7 +626,19 @@ __global__ void axpby_kernel(T* __restrict__ y, const T* __restrict__ x,
                              T alpha, T beta, uint64_t n) {
     uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
     uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
-    for (uint64_t j = i; j < n; j += stride)
+    using V = T __attribute__((ext_vector_type(4)));
+    uint64_t nv = n / 4;
+    const V* xv = reinterpret_cast<const V*>(x);
+    V* yv = reinterpret_cast<V*>(y);
+    for (uint64_t j = i; j < nv; j += stride) {
+        V a = xv[j], b = yv[j], r;
+        r.x = alpha * a.x + beta * b.x;
+        r.y = alpha * a.y + beta * b.y;
+        r.z = alpha * a.z + beta * b.z;
+        r.w = alpha * a.w + beta * b.w;
+        yv[j] = r;
+    }
+    for (uint64_t j = 4 * nv + i; j < n; j += stride)
         y[j] = alpha * x[j] + beta * y[j];
 }
 
@@ -653,10 +665,28 @@ __global__ void add_kernel(T* __restrict__ d, const T* __restrict__ s_,
                            T scale, int unit, uint64_t n) {
     uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
     uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    using V = T __attribute__((ext_vector_type(4)));
+    uint64_t nv = n / 4;
+    const V* sv = reinterpret_cast<const V*>(s_);
+    V* dv = reinterpret_cast<V*>(d);
     if (unit) {
-        for (uint64_t j = i; j < n; j += stride) d[j] = d[j] + s_[j];
+        for (uint64_t j = i; j < nv; j += stride) {
+            V a = dv[j], b = sv[j], r;
+            r.x = a.x + b.x; r.y = a.y + b.y;
+            r.z = a.z + b.z; r.w = a.w + b.w;
+            dv[j] = r;
+        }
+        for (uint64_t j = 4 * nv + i; j < n; j += stride)
+            d[j] = d[j] + s_[j];
     } else {
-        for (uint64_t j = i; j < n; j += stride) d[j] = d[j] + scale * s_[j];
+        for (uint64_t j = i; j < nv; j += stride) {
+            V a = dv[j], b = sv[j], r;
+            r.x = a.x + scale * b.x; r.y = a.y + scale * b.y;
+            r.z = a.z + scale * b.z; r.w = a.w + scale * b.w;
+            dv[j] = r;
+        }
+        for (uint64_t j = 4 * nv + i; j < n; j += stride)
+            d[j] = d[j] + scale * s_[j];
     }
 }
 
@@ -685,7 +715,15 @@ template <typename T>
 __global__ void scale_kernel(T* __restrict__ a, T v, uint64_t n) {
     uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
     uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
-    for (uint64_t j = i; j < n; j += stride) a[j] = a[j] * v;
+    using V = T __attribute__((ext_vector_type(4)));
+    uint64_t nv = n / 4;
+    V* av = reinterpret_cast<V*>(a);
+    for (uint64_t j = i; j < nv; j += stride) {
+        V x = av[j];
+        x.x *= v; x.y *= v; x.z *= v; x.w *= v;
+        av[j] = x;
+    }
+    for (uint64_t j = 4 * nv + i; j < n; j += stride) a[j] = a[j] * v;
 }
 
 int launch_scale(void* a, double s_, uint64_t n, int dtype, hipStream_t s) {
