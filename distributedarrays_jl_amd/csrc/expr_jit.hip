@@ -141,7 +141,7 @@ static bool prog_heavy(const int32_t* prog, int plen) {
 }
 
 std::string gen_source(const int32_t* prog, int plen, int dtype, int nd,
-                       int nsrcs, bool strided) {
+                       int nsrcs, bool strided, int sunroll) {
     const char* tname = dtype == DA_F64 ? "double"
                         : dtype == DA_F32 ? "float" : "long long";
     bool i64 = dtype == DA_I64;
@@ -183,6 +183,43 @@ std::string gen_source(const int32_t* prog, int plen, int dtype, int nd,
             "  for (unsigned long long j = %d * nv + i; j < a.n; j += st)\n"
             "    dst[j] = evalx(a, j);\n"
             "}\n", u);
+        s += buf;
+    } else if (sunroll == 4) {
+        // dim-0 extent is a multiple of 4 (checked by the launcher), so
+        // every aligned group of 4 flat indices stays within one dim-0
+        // run: decode once, step each operand by its dim-0 stride
+        int K = nsrcs > 0 ? nsrcs : 1;
+        snprintf(buf, sizeof(buf),
+            "/*s4*/extern \"C\" __global__ void ejit(JArgs a) {\n"
+            "  unsigned long long i0 = (unsigned long long)blockIdx.x * "
+            "blockDim.x + threadIdx.x;\n"
+            "  unsigned long long gs = (unsigned long long)gridDim.x * "
+            "blockDim.x;\n"
+            "  %s* dst = (%s*)a.dst;\n"
+            "  unsigned long long ng = a.n / 4;\n"
+            "  for (unsigned long long g = i0; g < ng; g += gs) {\n"
+            "    unsigned long long j = 4 * g;\n"
+            "    unsigned int idx[%d]; unsigned int rem = (unsigned int)j;\n"
+            "    for (int d = 0; d < %d; ++d) { idx[d] = rem %% a.dims[d]; "
+            "rem /= a.dims[d]; }\n"
+            "    unsigned long long off[%d], o1[%d], o2[%d], o3[%d];\n"
+            "    for (int k = 0; k < %d; ++k) {\n"
+            "      off[k] = 0;\n"
+            "      for (int d = 0; d < %d; ++d) off[k] += "
+            "(unsigned long long)idx[d] * a.str[k][d];\n"
+            "      o1[k] = off[k] + a.str[k][0];\n"
+            "      o2[k] = o1[k] + a.str[k][0];\n"
+            "      o3[k] = o2[k] + a.str[k][0];\n"
+            "    }\n"
+            "    %s r0 = evalx(a, off);\n"
+            "    %s r1 = evalx(a, o1);\n"
+            "    %s r2 = evalx(a, o2);\n"
+            "    %s r3 = evalx(a, o3);\n"
+            "    dst[j] = r0; dst[j + 1] = r1;\n"
+            "    dst[j + 2] = r2; dst[j + 3] = r3;\n"
+            "  }\n"
+            "}\n", tname, tname, nd, nd, K, K, K, K, K, nd,
+            tname, tname, tname, tname);
         s += buf;
     } else {
         snprintf(buf, sizeof(buf),
@@ -292,7 +329,12 @@ int launch_expr_jit(const int32_t* prog, int plen, void* dst,
     if (e && e[0] == '0') return 1;
     if (g_jit_state < 0) return 1;   // earlier hard failure: interpreter
     bool strided = src_strides != nullptr;
-    std::string src = gen_source(prog, plen, dtype, nd, nsrcs, strided);
+    int sunroll = 1;
+    if (strided && !prog_heavy(prog, plen) && dst_dims &&
+        dst_dims[0] % 4 == 0 && n % 4 == 0)
+        sunroll = 4;
+    std::string src = gen_source(prog, plen, dtype, nd, nsrcs, strided,
+                                 sunroll);
     hipFunction_t fn;
     if (jit_get(src, &fn) != 0) {
         g_jit_state = -1;            // remember; interpreter from now on
@@ -316,7 +358,8 @@ int launch_expr_jit(const int32_t* prog, int plen, void* dst,
     void* cfg[] = {HIP_LAUNCH_PARAM_BUFFER_POINTER, &a,
                    HIP_LAUNCH_PARAM_BUFFER_SIZE, &asz,
                    HIP_LAUNCH_PARAM_END};
-    uint64_t work = strided ? n : n / 2 + 1;
+    uint64_t work = strided ? (sunroll == 4 ? n / 4 + 1 : n)
+                            : n / 2 + 1;
     uint64_t b = (work + 255) / 256;
     if (b > 8192) b = 8192;   // 1024 workgroups/XCD fills the chip
     if (b == 0) b = 1;
@@ -340,7 +383,7 @@ extern "C" int dbg_expr_jit_compile(const int32_t* prog, int plen,
                                     int src_len) {
     using namespace da;
     std::string src = gen_source(prog, plen, dtype, nd, nsrcs,
-                                 strided != 0);
+                                 strided != 0, 1);
     if (src_out && src_len > 0) {
         strncpy(src_out, src.c_str(), src_len - 1);
         src_out[src_len - 1] = 0;
