@@ -330,8 +330,10 @@ int launch_expr_jit(const int32_t* prog, int plen, void* dst,
     if (g_jit_state < 0) return 1;   // earlier hard failure: interpreter
     bool strided = src_strides != nullptr;
     int sunroll = 1;
+    const char* s4 = getenv("DA_EJIT_S4");
     if (strided && !prog_heavy(prog, plen) && dst_dims &&
-        dst_dims[0] % 4 == 0 && n % 4 == 0)
+        dst_dims[0] % 4 == 0 && n % 4 == 0 &&
+        (!s4 || s4[0] != '0'))
         sunroll = 4;
     std::string src = gen_source(prog, plen, dtype, nd, nsrcs, strided,
                                  sunroll);
