@@ -329,11 +329,14 @@ int launch_expr_jit(const int32_t* prog, int plen, void* dst,
     if (e && e[0] == '0') return 1;
     if (g_jit_state < 0) return 1;   // earlier hard failure: interpreter
     bool strided = src_strides != nullptr;
+    // strided 4-wide measured a ~7% REGRESSION on the de-mean form
+    // (0.99 vs 0.93 ms same-box A/B: the three extra per-operand offset
+    // arrays outweigh the decode savings) — default off, kept for
+    // re-evaluation via DA_EJIT_S4=1
     int sunroll = 1;
     const char* s4 = getenv("DA_EJIT_S4");
-    if (strided && !prog_heavy(prog, plen) && dst_dims &&
-        dst_dims[0] % 4 == 0 && n % 4 == 0 &&
-        (!s4 || s4[0] != '0'))
+    if (s4 && s4[0] == '1' && strided && !prog_heavy(prog, plen) &&
+        dst_dims && dst_dims[0] % 4 == 0 && n % 4 == 0)
         sunroll = 4;
     std::string src = gen_source(prog, plen, dtype, nd, nsrcs, strided,
                                  sunroll);
