@@ -27,7 +27,7 @@ from .ops import (map_, dmap, map2_, elementwise, map2_scalar_, elementwise_scal
                   dsum_dims, dprod_dims, dmaximum_dims, dminimum_dims,
                   dmean_dims, dmatvec, dmatvec_adj, gather_box, map_general,
                   map_localparts, map_localparts_, redistribute,
-                  dmapslices, dppeval, dcast,
+                  dmapslices, dppeval, dcast, dreshape,
                   map2_general,
                   broadcast_fma_general, dsort, dtranspose, ddiag_lmul, ddiag_rmul,
                   dgetindex, dmul_)
@@ -43,7 +43,7 @@ __all__ = [
     "dreduce_dims", "dsum_dims", "dprod_dims", "dmaximum_dims",
     "dminimum_dims", "dmean_dims", "dmatvec", "dmatvec_adj",
     "map_localparts", "map_localparts_", "redistribute",
-    "dmapslices", "dppeval", "dcast",
+    "dmapslices", "dppeval", "dcast", "dreshape",
     "map2_general",
     "gather_box", "map_general", "broadcast_fma_general", "dsort",
     "dtranspose", "ddiag_lmul", "ddiag_rmul", "dgetindex", "dmul_",

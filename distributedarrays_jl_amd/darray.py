@@ -396,6 +396,19 @@ class DArray:
         mismatched cuts localize `other` onto self's boxes via the
         makelocal gather (collective — the branch is metadata-determined
         so every rank takes the same path)."""
+        if isinstance(other, np.ndarray):
+            # d == a::AbstractArray (darray.jl:403-414): each rank
+            # compares its localpart with the matching slice
+            if self.dims != tuple(other.shape):
+                return False
+            sl = tuple(slice(lo, hi) for lo, hi in self.lidx)
+            same = bool(np.array_equal(self.localpart(), other[sl]))
+            import torch.distributed as td
+            if self.nranks > 1 and td.is_initialized():
+                flags = [None] * self.nranks
+                td.all_gather_object(flags, same)
+                return all(flags)
+            return same
         if isinstance(other, DArray):
             if self.dims != other.dims:
                 return False

@@ -970,6 +970,81 @@ def _dest_boxes(dest):
     return boxes
 
 
+def dreshape(A, dims):
+    """reshape(A::DVector, dims) — darray.jl:612-636: each chunk of the
+    result pulls the source linear ranges backing its columns (the
+    reference fetches A[a:a+nr-1] per column).  Here: ONE collective
+    gather of each rank's bounding linear interval over xGMI, then a
+    strided on-device unpack (da_copy2d per dim-1 slab) — at most a
+    rows-span over-fetch for row-split grids, exact for column splits."""
+    import numpy as np
+    if A.ndims != 1:
+        raise DArrayError("dreshape: 1-D source only (as the reference)")
+    dims = tuple(int(x) for x in dims)
+    total = 1
+    for x in dims:
+        total *= x
+    if total != A.dims[0]:
+        raise DArrayError("dreshape: dimensions must be consistent "
+                          "with array size")
+    out = DArray(dims, A.dtype)
+    esz = DTYPE_SIZE[A.dtype]
+    nd = len(dims)
+
+    def lin(point):
+        a, mul = 0, 1
+        for d in range(nd):
+            a += point[d] * mul
+            mul *= dims[d]
+        return a
+
+    boxes = [None] * A.nranks
+    spans = {}
+    for c, r in enumerate(out.ranks):
+        box = out.idxs[c]
+        if geometry.nelems(box) == 0:
+            continue
+        lo = lin([b[0] for b in box])
+        hi = lin([b[1] - 1 for b in box]) + 1
+        boxes[r] = ((lo, hi),)
+        spans[r] = (lo, hi)
+    buf, _ = gather_box(A, boxes)
+    if buf is not None and out.lnumel:
+        g0 = spans[out.rank][0]
+        (r0, r1) = out.lidx[0]
+        nr = r1 - r0
+        outer = [out.lidx[d] for d in range(2, nd)]
+        o_sizes = [hi - lo for lo, hi in outer]
+        idx = [0] * len(o_sizes)
+        lshape = out.lshape
+        while True:
+            point = [r0, out.lidx[1][0] if nd > 1 else 0] + \
+                [outer[k][0] + idx[k] for k in range(len(o_sizes))]
+            src_off = lin(point[:nd]) - g0
+            dst_off = 0
+            mul = 1
+            for d in range(2, nd):
+                dst_off += idx[d - 2] * mul * lshape[0] * lshape[1]
+                mul *= lshape[d]
+            ncols = lshape[1] if nd > 1 else 1
+            _copy2d(out.at_byte(dst_off * esz), lshape[0] * esz,
+                    ctypes.c_void_p(buf.p.value + src_off * esz),
+                    dims[0] * esz, nr * esz, ncols)
+            k = 0
+            while k < len(o_sizes):
+                idx[k] += 1
+                if idx[k] < o_sizes[k]:
+                    break
+                idx[k] = 0
+                k += 1
+            if not o_sizes or k == len(o_sizes):
+                break
+        check(lib.da_synchronize())
+    if buf is not None:
+        buf.free()
+    return out
+
+
 def redistribute(D, dist):
     """A new DArray with the same global content on a different chunk
     grid — the reference's re-distribution constructor pattern

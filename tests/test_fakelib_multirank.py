@@ -733,3 +733,33 @@ def _scenario_outer_broadcast(rank, world, dja):
 @pytest.mark.timeout(420)
 def test_world2_outer_broadcast(tmp_path):
     _spawn(tmp_path, 2, "outer_broadcast")
+
+
+def _scenario_reshape(rank, world, dja):
+    """reshape(DVector, dims) — darray.jl:612-636 — plus == against a
+    plain host array (darray.jl:403-414)."""
+    n = 24 * world
+    g = _global_f64(n, 80)
+    D = _slice_set(dja.DArray((n,), "f64"), g)
+    # 1-D -> 2-D (column split: exact gather)
+    R = dja.dreshape(D, (6, n // 6))
+    ref = g.reshape((6, n // 6), order="F")
+    assert np.array_equal(R.collect(), ref)
+    assert (R == ref) is True
+    assert (R == (ref + 1e-12)) is False
+    # 1-D -> 3-D
+    R3 = dja.dreshape(D, (4, 3, n // 12))
+    assert np.array_equal(R3.collect(),
+                          g.reshape((4, 3, n // 12), order="F"))
+    # 1-D -> 1-D (identity relayout)
+    R1 = dja.dreshape(D, (n,))
+    assert np.array_equal(R1.collect(), g)
+    assert (D == g) is True
+    for d in (D, R, R3, R1):
+        d.close()
+
+
+@pytest.mark.timeout(420)
+@pytest.mark.parametrize("world", [2, 4])
+def test_reshape(tmp_path, world):
+    _spawn(tmp_path, world, "reshape")

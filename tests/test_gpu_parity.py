@@ -699,3 +699,20 @@ def test_index_sugar(dja):
     assert np.array_equal(M[4, 0:8], m[4, 0:8])
     assert np.array_equal(M[:, 5], m[:, 5])
     M.close()
+
+
+def test_dreshape_and_eq_array(dja):
+    """reshape (darray.jl:612-636) and == vs a host array
+    (darray.jl:403-414)."""
+    n = 840
+    x = philox.fill_uniform_f64(n, 60)
+    d = dja.distribute(x)
+    R = dja.dreshape(d, (21, 40))
+    ref = x.reshape((21, 40), order="F")
+    assert np.array_equal(R.collect(), ref)
+    assert (R == ref) is True
+    R3 = dja.dreshape(d, (7, 4, 30))
+    assert np.array_equal(R3.collect(), x.reshape((7, 4, 30), order="F"))
+    assert (d == x) is True
+    assert (d == (x * 2)) is False
+    R.close(); R3.close(); d.close()
